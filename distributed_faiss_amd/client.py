@@ -1,0 +1,325 @@
+# IndexClient: shard fan-out + top-k merge.
+#
+# Keeps the reference's client surface (distributed_faiss/client.py:57-345):
+# create/load/save/drop index, add_index_data round-robin placement
+# (client.py:174-192), search fan-out + merge (client.py:200-210,265-310),
+# search_with_filter (client.py:213-263), get_state aggregation, get_ntotal
+# sum, get_centroids, set_nprobe, get_ids. Differences from the reference:
+#   * shards are in-process IndexServer objects (or, multi-GPU, one rank per
+#     GPU via distributed_faiss_amd.dist) instead of pickle-TCP rpc.Client
+#     proxies — the RPC layer is retired for the intra-node path
+#     (BASELINE.json north_star; reference rpc.py out of scope per
+#     SURVEY.md §2).
+#   * the merge is a numpy restatement of ResultHeap
+#     (faiss.float_maxheap_array_t, client.py:29-54) pinned by the
+#     reference's own merge KAT (tests/test_integration.py:181-203, ported
+#     into tests/test_merge.py).
+#
+# Merge semantics mirrored EXACTLY, including reference quirk 2
+# (SURVEY.md §2): for metric "dot" the client pushes -D into the
+# min-oriented merge and NEVER flips the sign back — returned scores for
+# dot are negated. Unfilled shard slots arrive as D=+FLT_MAX (L2) /
+# -FLT_MAX (dot) with id -1 and meta None, and lose the merge naturally.
+
+import itertools
+import logging
+import os
+import random
+import time
+from multiprocessing.dummy import Pool as ThreadPool
+from typing import List, Optional, Tuple
+
+import numpy as np
+
+from .index_cfg import IndexCfg
+from .index_state import IndexState
+from .server import IndexServer, resolve_inproc_server
+
+logger = logging.getLogger()
+
+
+class ResultHeap:
+    """Merge accumulator, numpy restatement of reference client.py:29-54.
+
+    Keeps the k SMALLEST values per query (the caller negates for
+    maximize). Ties break toward the earlier-added synthetic id, which is
+    deterministic (faiss's heap order on ties is unspecified; the KAT has
+    no ties).
+    """
+
+    def __init__(self, nq, k):
+        self.nq, self.k = nq, k
+        self._vals = []   # list of (nq, k) fp32
+        self._ids = []    # list of (nq, k) int64
+        self.D = None
+        self.I = None
+
+    def add_result(self, D, I):
+        assert D.shape == (self.nq, self.k)
+        assert I.shape == (self.nq, self.k)
+        self._vals.append(np.asarray(D, dtype=np.float32))
+        self._ids.append(np.asarray(I, dtype=np.int64))
+
+    def finalize(self):
+        allD = np.concatenate(self._vals, axis=1)
+        allI = np.concatenate(self._ids, axis=1)
+        self.D = np.empty((self.nq, self.k), dtype=np.float32)
+        self.I = np.empty((self.nq, self.k), dtype=np.int64)
+        for i in range(self.nq):
+            order = np.lexsort((allI[i], allD[i]))[: self.k]
+            self.D[i] = allD[i, order]
+            self.I[i] = allI[i, order]
+
+
+class IndexClient:
+    """Manages a set of shard sub-indexes; searches merge client-side."""
+
+    def __init__(self, server_list_path=None, cfg_path: Optional[str] = None, servers=None):
+        """Either pass `servers` (list of in-process IndexServer) or
+        `server_list_path` (reference format: first line = count, then
+        host,port lines — resolved against the in-process port registry)."""
+        if servers is not None:
+            self.sub_indexes: List[IndexServer] = list(servers)
+        else:
+            machine_ports = IndexClient.read_server_list(server_list_path)
+            self.sub_indexes = [resolve_inproc_server(port) for _host, port in machine_ports]
+        self.num_indexes = len(self.sub_indexes)
+
+        index_ranks = [idx.get_rank() for idx in self.sub_indexes]
+        self.index_rank_to_id = {
+            index_rank: index_id for index_id, index_rank in enumerate(index_ranks)
+        }
+
+        self.pool = ThreadPool(self.num_indexes)
+        self.verbose = False
+        self.cur_server_ids = {}
+        random.seed(time.time())
+        self.cfg = IndexCfg.from_json(cfg_path) if cfg_path is not None else None
+
+    @staticmethod
+    def read_server_list(
+        server_list_path,
+        initial_timeout=0.1,
+        backoff_factor=1.5,
+        total_max_timeout=7200,
+    ) -> List[Tuple[str, int]]:
+        # reference client.py:87-120 restated, including the no-success-break
+        # spin (quirk 8): loop until the timeout accumulator trips, then
+        # assert the count.
+        time_waited = 0
+        while True:
+            with open(server_list_path) as f:
+                res = []
+                num_servers = 0
+                for idx, line in enumerate(f):
+                    if idx == 0:
+                        num_servers = int(line)
+                        continue
+                    res.append((str(line.split(",")[0]), int(line.split(",")[1])))
+
+            msg = f"{num_servers} != {len(res)} in server list {server_list_path}."
+            if num_servers != len(res):
+                time.sleep(initial_timeout)
+            if time_waited + initial_timeout >= total_max_timeout:
+                break
+            time_waited += initial_timeout
+            initial_timeout *= backoff_factor
+
+        assert num_servers == len(res), (
+            msg + f" Timed out after waiting {round(time_waited * 100) / 100} seconds"
+        )
+        return res
+
+    # -- index management --------------------------------------------------
+
+    def create_index(self, index_id: str, cfg: Optional[IndexCfg] = None):
+        if cfg is not None:
+            self.cfg = cfg
+        if self.cfg is None:
+            self.cfg = IndexCfg()
+        return self.pool.map(lambda idx: idx.create_index(index_id, self.cfg), self.sub_indexes)
+
+    def load_index(
+        self, index_id: str, cfg: Optional[IndexCfg] = None, force_reload: bool = True
+    ) -> bool:
+        def setup_cfg(cfg):
+            if cfg is None:
+                config_paths = self.pool.map(
+                    lambda idx: idx.get_config_path(index_id), self.sub_indexes
+                )
+                if len(config_paths) > 0 and os.path.isfile(config_paths[0]):
+                    cfg = IndexCfg.from_json(config_paths[0])
+                else:
+                    cfg = IndexCfg()
+            return cfg
+
+        if force_reload:
+            self.pool.map(lambda idx: idx.drop_index(index_id), self.sub_indexes)
+        all_loaded = self.pool.map(lambda idx: idx.load_index(index_id, cfg), self.sub_indexes)
+        self.cfg = setup_cfg(cfg)
+        if all(all_loaded):
+            return True
+        if any(all_loaded):
+            logger.warning(f"Some server nodes can't load index: {all_loaded}")
+        return False
+
+    def save_index(self, index_id: str):
+        self.pool.map(lambda idx: idx.save_index(index_id), self.sub_indexes)
+
+    def drop_index(self, index_id: str):
+        self.pool.map(lambda idx: idx.drop_index(index_id), self.sub_indexes)
+
+    # -- data path ---------------------------------------------------------
+
+    def add_index_data(
+        self,
+        index_id: str,
+        embeddings: np.ndarray,
+        metadata: Optional[List[object]] = None,
+        train_async_if_triggered: bool = True,
+    ) -> None:
+        # round-robin placement, random start (reference client.py:174-192)
+        if index_id not in self.cur_server_ids:
+            self.cur_server_ids[index_id] = random.randint(0, self.num_indexes - 1)
+        cur_server_id = self.cur_server_ids[index_id]
+        self.sub_indexes[cur_server_id].add_index_data(
+            index_id, embeddings, metadata, train_async_if_triggered
+        )
+        self.cur_server_ids[index_id] = (self.cur_server_ids[index_id] + 1) % self.num_indexes
+
+    def sync_train(self, index_id: str) -> None:
+        self.pool.map(lambda idx: idx.sync_train(index_id), self.sub_indexes)
+
+    def async_train(self, index_id: str):
+        # reference quirk 4: client.async_train == sync_train (client.py:197-198)
+        self.pool.map(lambda idx: idx.sync_train(index_id), self.sub_indexes)
+
+    def search(
+        self, query, topk: int, index_id: str, return_embeddings: bool = False
+    ) -> Tuple[np.ndarray, List]:
+        q_size = query.shape[0]
+        maximize_metric: bool = self.cfg.metric == "dot"
+        results = self.pool.imap(
+            lambda idx: idx.search(index_id, query, topk, return_embeddings), self.sub_indexes
+        )
+        return self._aggregate_results(results, topk, q_size, maximize_metric, return_embeddings)
+
+    def search_with_filter(
+        self,
+        query: np.ndarray,
+        top_k: int,
+        index_id: str,
+        filter_pos: int = -1,
+        filter_value=None,
+    ) -> Tuple[np.ndarray, List[List[object]]]:
+        # reference client.py:213-263: over-fetch x3, host-side post-filter
+        filter_top_factor = 3
+        actual_top_k = filter_top_factor * top_k if filter_pos >= 0 else top_k
+        (scores, meta) = self.search(query, actual_top_k, index_id)
+        if filter_pos < 0:
+            return scores, meta
+
+        def _do_filter(scores, results_meta):
+            re_query_ids = []
+            new_results = []
+            new_scores = []
+            for i, meta_list in enumerate(results_meta):
+                sample_filtered_meta = []
+                sample_filtered_scores = []
+                for j, m in enumerate(meta_list):
+                    if not m:
+                        continue
+                    if len(m) > filter_pos and m[filter_pos] != filter_value:
+                        sample_filtered_meta.append(m)
+                        sample_filtered_scores.append(scores[i, j])
+                    if len(sample_filtered_meta) >= top_k:
+                        break
+                if len(sample_filtered_meta) < top_k:
+                    re_query_ids.append(i)
+                new_results.append(sample_filtered_meta)
+                new_scores.append(
+                    np.concatenate([s.reshape(-1, 1) for s in sample_filtered_scores], axis=0)
+                    if sample_filtered_scores
+                    else np.empty((0, 1), dtype=np.float32)
+                )
+            return new_scores, new_results, re_query_ids
+
+        new_scores, new_results_meta, _re_query_ids = _do_filter(scores, meta)
+        return new_scores, new_results_meta
+
+    @staticmethod
+    def _aggregate_results(
+        results,
+        topk: int,
+        q_size: int,
+        maximize_metric: bool,
+        return_embeddings: bool,
+    ):
+        # reference client.py:265-310 restated: extend a flat metadata list,
+        # re-index candidates with synthetic ids, push (-D for dot) into the
+        # min-merge, map winning synthetic ids back to metadata.
+        meta = []
+        embs = []
+        cur_idx = 0
+
+        def to_matrix(l, n):
+            return [l[i : i + n] for i in range(0, len(l), n)]
+
+        res_heap = ResultHeap(q_size, topk)
+        for DI, MetaI, e in results:
+            merged_meta = list(itertools.chain(*MetaI))
+            meta.extend(merged_meta)
+            if return_embeddings:
+                merged_embs = list(itertools.chain(*e))
+                embs.extend(merged_embs)
+            Ii = np.reshape(np.arange(cur_idx, cur_idx + q_size * topk), (q_size, topk))
+            if maximize_metric:
+                res_heap.add_result(-DI, Ii)
+            else:
+                res_heap.add_result(DI, Ii)
+            cur_idx += q_size * topk
+        res_heap.finalize()
+        ids = np.reshape(res_heap.I, (-1,)).tolist()
+        selected_meta = [meta[i] for i in ids]
+        if return_embeddings:
+            selected_embs = [embs[i] for i in ids]
+        return (
+            (
+                res_heap.D,
+                to_matrix(selected_meta, res_heap.D.shape[1]),
+                to_matrix(selected_embs, res_heap.D.shape[1]),
+            )
+            if return_embeddings
+            else (res_heap.D, to_matrix(selected_meta, res_heap.D.shape[1]))
+        )
+
+    # -- introspection / knobs --------------------------------------------
+
+    def get_centroids(self, index_id: str):
+        return self.pool.map(lambda idx: idx.get_centroids(index_id), self.sub_indexes)
+
+    def set_nprobe(self, index_id: str, nprobe: int):
+        return self.pool.map(lambda idx: idx.set_nprobe(index_id, nprobe), self.sub_indexes)
+
+    def get_state(self, index_id: str) -> IndexState:
+        states = self.pool.map(lambda idx: idx.get_state(index_id), self.sub_indexes)
+        return IndexState.get_aggregated_states(states)
+
+    def add_buffer_to_index(self, index_id: str):
+        self.pool.map(lambda idx: idx.add_buffer_to_index(index_id), self.sub_indexes)
+
+    def get_ntotal(self, index_id: str) -> int:
+        return sum(self.pool.map(lambda idx: idx.get_ntotal(index_id), self.sub_indexes))
+
+    def get_ids(self, index_id: str) -> set:
+        id_set_list = self.pool.map(lambda idx: idx.get_ids(index_id), self.sub_indexes)
+        return set().union(*id_set_list)
+
+    def set_omp_num_threads(self, num_threads: int) -> None:
+        self.pool.map(lambda idx: idx.set_omp_num_threads(num_threads), self.sub_indexes)
+
+    def close(self):
+        pass  # no sockets to close in-process
+
+    def get_num_servers(self):
+        return self.num_indexes
