@@ -1,0 +1,1100 @@
+// dfann — MI355X-native ANN engine: host runtime + C-ABI (include/dfann.h).
+//
+// Self-contained HIP/C++ shared library: no torch, no BLAS — every FLOP of
+// the hot path runs in the hand-written gfx950 kernels of kernels.hip.
+// Python binds via ctypes (distributed_faiss_amd/hip_engine.py).
+//
+// Index model:
+//  * trained artifacts: coarse centroids (+norms), PQ codebooks / SQ ranges
+//  * codes live twice: an arrival-order staging arena (append target) and
+//    a CSR image grouped by inverted list (search layout), rebuilt lazily
+//    ("finalize") via a host counting sort + device gather — the stable
+//    sort preserves arrival order inside each list, so CSR position order
+//    == ascending arrival id, which the scan's tie-break relies on.
+//  * ids are implicit arrival positions (reference quirk).
+
+#include "kernels.hip"
+
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+#include <stdexcept>
+#include <algorithm>
+
+#include "../../include/dfann.h"
+
+// ---------------------------------------------------------------------------
+// error plumbing
+// ---------------------------------------------------------------------------
+
+static thread_local std::string g_err;
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess) {                                                    \
+      throw std::runtime_error(std::string("HIP error: ") +                    \
+                               hipGetErrorString(_e) + " at " #expr);          \
+    }                                                                          \
+  } while (0)
+
+extern "C" const char *dfann_last_error(void) { return g_err.c_str(); }
+
+// ---------------------------------------------------------------------------
+// tiny JSON (flat dict of scalars; produced by our own Python side)
+// ---------------------------------------------------------------------------
+
+static bool json_find(const std::string &js, const char *key, std::string &out) {
+  std::string pat = std::string("\"") + key + "\"";
+  size_t p = js.find(pat);
+  if (p == std::string::npos) return false;
+  p = js.find(':', p + pat.size());
+  if (p == std::string::npos) return false;
+  ++p;
+  while (p < js.size() && (js[p] == ' ' || js[p] == '\t')) ++p;
+  if (p >= js.size()) return false;
+  if (js[p] == '"') {
+    size_t e = js.find('"', p + 1);
+    out = js.substr(p + 1, e - p - 1);
+  } else {
+    size_t e = p;
+    while (e < js.size() && (isdigit(js[e]) || js[e] == '-' || js[e] == '.' ||
+                             js[e] == 'e' || js[e] == 'E' || js[e] == '+'))
+      ++e;
+    out = js.substr(p, e - p);
+  }
+  return true;
+}
+
+static long long json_int(const std::string &js, const char *key, long long dflt) {
+  std::string v;
+  if (!json_find(js, key, v) || v.empty()) return dflt;
+  return atoll(v.c_str());
+}
+
+static std::string json_str(const std::string &js, const char *key,
+                            const char *dflt) {
+  std::string v;
+  if (!json_find(js, key, v)) return dflt;
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// device buffer
+// ---------------------------------------------------------------------------
+
+struct DevBuf {
+  void *p = nullptr;
+  size_t cap = 0;
+  void ensure(size_t bytes) {
+    if (bytes <= cap) return;
+    void *np = nullptr;
+    HIP_CHECK(hipMalloc(&np, bytes));
+    if (p) HIP_CHECK(hipFree(p));
+    p = np;
+    cap = bytes;
+  }
+  // grow preserving `keep` bytes of content
+  void grow_keep(size_t bytes, size_t keep) {
+    if (bytes <= cap) return;
+    size_t nb = std::max(bytes, cap * 2);
+    void *np = nullptr;
+    HIP_CHECK(hipMalloc(&np, nb));
+    if (p && keep) HIP_CHECK(hipMemcpy(np, p, keep, hipMemcpyDeviceToDevice));
+    if (p) HIP_CHECK(hipFree(p));
+    p = np;
+    cap = nb;
+  }
+  void free() {
+    if (p) hipFree(p);
+    p = nullptr;
+    cap = 0;
+  }
+  template <typename T> T *as() { return reinterpret_cast<T *>(p); }
+};
+
+// ---------------------------------------------------------------------------
+// splitmix64 (identical to oracle/core.py)
+// ---------------------------------------------------------------------------
+
+struct SplitMix64 {
+  uint64_t x;
+  explicit SplitMix64(uint64_t seed) : x(seed) {}
+  uint64_t next() {
+    uint64_t z = (x += 0x9E3779B97F4A7C15ULL);
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBULL;
+    return z ^ (z >> 31);
+  }
+};
+
+// first k of a partial Fisher-Yates over range(n) (== oracle
+// partial_shuffle_indices)
+static std::vector<int64_t> pick_init(int64_t n, int64_t k, uint64_t seed) {
+  k = std::min(k, n);
+  std::vector<int64_t> idx(n);
+  for (int64_t i = 0; i < n; ++i) idx[i] = i;
+  SplitMix64 rng(seed);
+  for (int64_t i = 0; i < k; ++i) {
+    int64_t j = i + (int64_t)(rng.next() % (uint64_t)(n - i));
+    std::swap(idx[i], idx[j]);
+  }
+  idx.resize(k);
+  return idx;
+}
+
+// ---------------------------------------------------------------------------
+// index object
+// ---------------------------------------------------------------------------
+
+enum IdxType { T_FLAT = 0, T_IVFFLAT = 1, T_IVFPQ = 2, T_IVFSQ = 3 };
+enum { M_IP = 0, M_L2 = 1 };
+
+static int round16(int b) { return (b + 15) & ~15; }
+
+struct TimingEv {
+  hipEvent_t a, b;
+};
+
+struct dfann_index {
+  std::string spec_json;
+  int type = T_FLAT;
+  int metric = M_IP;
+  int d = 0, nlist = 0, m = 0, nbits = 8, dsub = 0, nprobe = 1;
+  bool sq8 = false;  // ivfsq: true = 8bit, false = fp16
+  uint64_t seed = 1234;
+  bool trained = false;
+  int64_t ntotal = 0;
+  int code_bytes = 0, stride = 0;
+
+  DevBuf centroids, cnorm, codebooks, sq_vmin, sq_vdiff, sq_scale;
+  // staging (arrival order)
+  DevBuf st_codes;
+  std::vector<int32_t> h_assign;
+  // CSR
+  DevBuf cr_codes, cr_ids, cr_off, id2pos;
+  std::vector<int64_t> h_off;
+  bool dirty = false;
+  // flat arena
+  DevBuf flat;
+  // workspace
+  DevBuf ws1, ws2, ws3, ws4, ws5;
+
+  // timing
+  bool timing = false;
+  std::vector<TimingEv> ev_scan, ev_gemm, ev_merge;
+  int64_t scan_rows = 0, scan_bytes = 0, gemm_flops = 0;
+
+  ~dfann_index() {
+    for (auto &v : {ev_scan, ev_gemm, ev_merge})
+      for (auto &e : v) {
+        hipEventDestroy(e.a);
+        hipEventDestroy(e.b);
+      }
+  }
+
+  TimingEv ev_begin(hipStream_t s) {
+    TimingEv e;
+    HIP_CHECK(hipEventCreate(&e.a));
+    HIP_CHECK(hipEventCreate(&e.b));
+    HIP_CHECK(hipEventRecord(e.a, s));
+    return e;
+  }
+  void ev_end(TimingEv e, hipStream_t s, std::vector<TimingEv> &dst) {
+    HIP_CHECK(hipEventRecord(e.b, s));
+    dst.push_back(e);
+  }
+};
+
+// ---------------------------------------------------------------------------
+// launch helpers
+// ---------------------------------------------------------------------------
+
+static dim3 grid1d(int64_t total, int block = 256, int64_t cap = 65535LL * 8) {
+  int64_t g = (total + block - 1) / block;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return dim3((unsigned)g);
+}
+
+// C[i][j] keys for rows of A vs rows of B under `metric`.
+// mode: 0 -ip, 1 bn-2ip, 2 qn+bn-2ip (needs qn). Result in `keys` (rows x N).
+static void gemm_keys(dfann_index *h, const float *A, int64_t Mrows,
+                      const float *B, int64_t N, int K, const float *bn,
+                      const float *qn, int mode, float *keys,
+                      hipStream_t stream) {
+  dim3 g((unsigned)((N + GT - 1) / GT), (unsigned)((Mrows + GT - 1) / GT));
+  TimingEv e;
+  if (h && h->timing) e = h->ev_begin(stream);
+  hipLaunchKernelGGL(k_gemm_nt, g, dim3(256), 0, stream, A, B, keys, (int)Mrows,
+                     (int)N, K, K, K, (int)N);
+  hipLaunchKernelGGL(k_score_key, grid1d(Mrows * N), dim3(256), 0, stream, keys,
+                     Mrows, N, N, qn, bn, mode);
+  if (h && h->timing) {
+    h->ev_end(e, stream, h->ev_gemm);
+    h->gemm_flops += 2LL * Mrows * N * K;
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+static void rownorms(const float *x, int64_t n, int d, float *out,
+                     hipStream_t stream) {
+  if (n == 0) return;
+  hipLaunchKernelGGL(k_rownorm, dim3((unsigned)n), dim3(256), 0, stream, x, n, d,
+                     out);
+  HIP_CHECK(hipGetLastError());
+}
+
+// nearest-centroid assignment of (n x d) rows against h->centroids
+static void assign_rows(dfann_index *h, const float *x, int64_t n,
+                        int32_t *assign_dev, hipStream_t stream) {
+  int nlist = h->nlist;
+  // chunk points so the key matrix stays <= ~512 MB
+  int64_t chunk = std::max<int64_t>(1, (512LL << 20) / ((int64_t)nlist * 4));
+  chunk = std::min<int64_t>(chunk, n);
+  h->ws1.ensure((size_t)chunk * nlist * 4);
+  float *keys = h->ws1.as<float>();
+  h->ws2.ensure((size_t)n * 4);
+  float *bestv = h->ws2.as<float>();
+  hipLaunchKernelGGL(k_assign_init, grid1d(n), dim3(256), 0, stream, bestv,
+                     assign_dev, n);
+  for (int64_t s = 0; s < n; s += chunk) {
+    int64_t c = std::min(chunk, n - s);
+    gemm_keys(h, x + s * h->d, c, h->centroids.as<float>(), nlist, h->d,
+              h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, keys,
+              stream);
+    hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream, keys, c,
+                       (long long)nlist, (long long)nlist, 0, bestv,
+                       assign_dev + s);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// k-means (restating faiss Clustering; deviations shared with the oracle —
+// oracle/__init__.py header): niter 25, strided subsample to k*256, seeded
+// partial-FY init, metric-driven assignment, mean update, deterministic
+// largest-donor empty-cluster split.
+// ---------------------------------------------------------------------------
+
+static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
+                          int d, int metric, uint64_t seed, float *cent_out,
+                          hipStream_t stream) {
+  const int NITER = 25;
+  int64_t cap = (int64_t)kcent * 256;
+  DevBuf xt_buf;
+  const float *xt = x;
+  int64_t nt = n;
+  if (n > cap) {
+    xt_buf.ensure((size_t)cap * d * 4);
+    hipLaunchKernelGGL(k_gather_strided, grid1d(cap * d), dim3(256), 0, stream,
+                       x, n, cap, d, xt_buf.as<float>());
+    xt = xt_buf.as<float>();
+    nt = cap;
+  }
+  if (nt < kcent) throw std::runtime_error("kmeans: n < k");
+
+  // init centroids
+  auto idx = pick_init(nt, kcent, seed);
+  DevBuf idx_buf;
+  idx_buf.ensure(idx.size() * 8);
+  HIP_CHECK(hipMemcpyAsync(idx_buf.p, idx.data(), idx.size() * 8,
+                           hipMemcpyHostToDevice, stream));
+  hipLaunchKernelGGL(k_gather_rows, grid1d((int64_t)kcent * d), dim3(256), 0,
+                     stream, xt, idx_buf.as<int64_t>(), (long long)kcent, d,
+                     cent_out);
+
+  DevBuf cn, asg, bestv, keys, sums, counts;
+  cn.ensure((size_t)kcent * 4);
+  asg.ensure((size_t)nt * 4);
+  bestv.ensure((size_t)nt * 4);
+  sums.ensure((size_t)kcent * d * 4);
+  counts.ensure((size_t)kcent * 4);
+  int64_t chunk = std::max<int64_t>(1, (512LL << 20) / ((int64_t)kcent * 4));
+  chunk = std::min(chunk, nt);
+  keys.ensure((size_t)chunk * kcent * 4);
+  std::vector<int> h_counts(kcent);
+  std::vector<float> h_cent;
+
+  for (int it = 0; it < NITER; ++it) {
+    rownorms(cent_out, kcent, d, cn.as<float>(), stream);
+    hipLaunchKernelGGL(k_assign_init, grid1d(nt), dim3(256), 0, stream,
+                       bestv.as<float>(), asg.as<int>(), nt);
+    for (int64_t s = 0; s < nt; s += chunk) {
+      int64_t c = std::min(chunk, nt - s);
+      gemm_keys(nullptr, xt + s * d, c, cent_out, kcent, d, cn.as<float>(),
+                nullptr, metric == M_IP ? 0 : 1, keys.as<float>(), stream);
+      hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream,
+                         keys.as<float>(), c, (long long)kcent,
+                         (long long)kcent, 0, bestv.as<float>(),
+                         asg.as<int>() + s);
+    }
+    HIP_CHECK(hipMemsetAsync(sums.p, 0, (size_t)kcent * d * 4, stream));
+    HIP_CHECK(hipMemsetAsync(counts.p, 0, (size_t)kcent * 4, stream));
+    hipLaunchKernelGGL(k_centroid_accum, grid1d(nt * d), dim3(256), 0, stream,
+                       xt, asg.as<int>(), nt, d, sums.as<float>(),
+                       counts.as<int>());
+    hipLaunchKernelGGL(k_centroid_div, grid1d((int64_t)kcent * d), dim3(256), 0,
+                       stream, cent_out, sums.as<float>(), counts.as<int>(),
+                       (long long)kcent, d);
+    // empty clusters: deterministic largest-donor split (== oracle)
+    HIP_CHECK(hipMemcpyAsync(h_counts.data(), counts.p, (size_t)kcent * 4,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    bool any_empty = false;
+    for (int c = 0; c < kcent; ++c)
+      if (h_counts[c] == 0) { any_empty = true; break; }
+    if (any_empty) {
+      h_cent.resize((size_t)kcent * d);
+      HIP_CHECK(hipMemcpy(h_cent.data(), cent_out, (size_t)kcent * d * 4,
+                          hipMemcpyDeviceToHost));
+      std::vector<int> cw(h_counts.begin(), h_counts.end());
+      const float eps = 1.0f / 1024.0f;
+      for (int ci = 0; ci < kcent; ++ci) {
+        if (h_counts[ci] != 0) continue;
+        int cj = (int)(std::max_element(cw.begin(), cw.end()) - cw.begin());
+        for (int t = 0; t < d; ++t) {
+          float v = h_cent[(size_t)cj * d + t];
+          h_cent[(size_t)ci * d + t] = v * (1.0f + eps);
+          h_cent[(size_t)cj * d + t] = v * (1.0f - eps);
+        }
+        cw[ci] = cw[cj] / 2;
+        cw[cj] -= cw[cj] / 2;
+      }
+      HIP_CHECK(hipMemcpy(cent_out, h_cent.data(), (size_t)kcent * d * 4,
+                          hipMemcpyHostToDevice));
+    }
+  }
+  xt_buf.free();
+}
+
+// ---------------------------------------------------------------------------
+// create / train / add / finalize
+// ---------------------------------------------------------------------------
+
+static dfann_index *create_from_spec(const std::string &js) {
+  auto *h = new dfann_index();
+  h->spec_json = js;
+  std::string t = json_str(js, "type", "");
+  if (t == "flat") h->type = T_FLAT;
+  else if (t == "ivf_flat") h->type = T_IVFFLAT;
+  else if (t == "ivfpq") h->type = T_IVFPQ;
+  else if (t == "ivfsq") h->type = T_IVFSQ;
+  else { delete h; throw std::runtime_error("unknown index type '" + t + "'"); }
+  h->d = (int)json_int(js, "dim", 0);
+  h->metric = (int)json_int(js, "metric", M_IP);
+  h->nlist = (int)json_int(js, "nlist", 0);
+  h->m = (int)json_int(js, "m", 0);
+  h->nbits = (int)json_int(js, "nbits", 8);
+  h->nprobe = (int)json_int(js, "nprobe", 1);
+  h->seed = (uint64_t)json_int(js, "seed", 1234);
+  h->sq8 = json_str(js, "sq_type", "fp16") == "8bit";
+  if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
+  if (h->type != T_FLAT && h->nlist <= 0) {
+    delete h;
+    throw std::runtime_error("bad nlist");
+  }
+  if (h->type == T_IVFPQ) {
+    if (h->nbits != 8) { delete h; throw std::runtime_error("only nbits=8"); }
+    if (h->m <= 0 || h->d % h->m) {
+      delete h;
+      throw std::runtime_error("bad m (dim % m != 0)");
+    }
+    h->dsub = h->d / h->m;
+    h->code_bytes = h->m;
+  } else if (h->type == T_IVFSQ) {
+    h->code_bytes = h->sq8 ? h->d : 2 * h->d;
+  } else if (h->type == T_IVFFLAT) {
+    h->code_bytes = 4 * h->d;
+  } else {
+    h->code_bytes = 4 * h->d;
+    h->trained = true;  // flat needs no training
+  }
+  h->stride = round16(h->code_bytes);
+  return h;
+}
+
+static void train_impl(dfann_index *h, int64_t n, const float *x,
+                       hipStream_t stream) {
+  if (h->type == T_FLAT) { h->trained = true; return; }
+  if (h->trained) return;  // faiss semantics: re-train of trained is a no-op here
+  h->centroids.ensure((size_t)h->nlist * h->d * 4);
+  h->cnorm.ensure((size_t)h->nlist * 4);
+  kmeans_device(h, x, n, h->nlist, h->d, h->metric, h->seed,
+                h->centroids.as<float>(), stream);
+  rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(),
+           stream);
+  if (h->type == T_IVFPQ || (h->type == T_IVFSQ && h->sq8)) {
+    DevBuf asg, resid;
+    asg.ensure((size_t)n * 4);
+    resid.ensure((size_t)n * h->d * 4);
+    assign_rows(h, x, n, asg.as<int>(), stream);
+    hipLaunchKernelGGL(k_residual, grid1d(n * h->d), dim3(256), 0, stream, x,
+                       h->centroids.as<float>(), asg.as<int>(), n, h->d,
+                       resid.as<float>());
+    if (h->type == T_IVFPQ) {
+      h->codebooks.ensure((size_t)h->m * 256 * h->dsub * 4);
+      DevBuf sub;
+      sub.ensure((size_t)n * h->dsub * 4);
+      for (int j = 0; j < h->m; ++j) {
+        hipLaunchKernelGGL(k_subspace_slice, grid1d(n * h->dsub), dim3(256), 0,
+                           stream, resid.as<float>(), n, h->d, j * h->dsub,
+                           h->dsub, sub.as<float>());
+        kmeans_device(h, sub.as<float>(), n, 256, h->dsub, M_L2,
+                      h->seed + 1 + j,
+                      h->codebooks.as<float>() + (size_t)j * 256 * h->dsub,
+                      stream);
+      }
+    } else {  // SQ8 ranges on residuals
+      DevBuf mn, mx;
+      mn.ensure((size_t)h->d * 4);
+      mx.ensure((size_t)h->d * 4);
+      h->sq_vmin.ensure((size_t)h->d * 4);
+      h->sq_vdiff.ensure((size_t)h->d * 4);
+      h->sq_scale.ensure((size_t)h->d * 4);
+      hipLaunchKernelGGL(k_minmax_init, grid1d(h->d), dim3(256), 0, stream,
+                         mn.as<unsigned>(), mx.as<unsigned>(), h->d);
+      hipLaunchKernelGGL(k_minmax_dims, grid1d(n * h->d), dim3(256), 0, stream,
+                         resid.as<float>(), n, h->d, mn.as<unsigned>(),
+                         mx.as<unsigned>());
+      hipLaunchKernelGGL(k_minmax_decode, grid1d(h->d), dim3(256), 0, stream,
+                         mn.as<unsigned>(), mx.as<unsigned>(), h->d,
+                         h->sq_vmin.as<float>(), h->sq_vdiff.as<float>(),
+                         h->sq_scale.as<float>());
+    }
+  }
+  HIP_CHECK(hipStreamSynchronize(stream));
+  h->trained = true;
+}
+
+static void add_impl(dfann_index *h, int64_t n, const float *x,
+                     hipStream_t stream) {
+  if (!h->trained) throw std::runtime_error("add on untrained index");
+  if (n == 0) return;
+  if (h->type == T_FLAT) {
+    h->flat.grow_keep(((size_t)h->ntotal + n) * h->d * 4,
+                      (size_t)h->ntotal * h->d * 4);
+    HIP_CHECK(hipMemcpyAsync(h->flat.as<float>() + (size_t)h->ntotal * h->d, x,
+                             (size_t)n * h->d * 4, hipMemcpyDeviceToDevice,
+                             stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    h->ntotal += n;
+    return;
+  }
+  DevBuf asg;
+  asg.ensure((size_t)n * 4);
+  assign_rows(h, x, n, asg.as<int>(), stream);
+  h->st_codes.grow_keep(((size_t)h->ntotal + n) * h->stride,
+                        (size_t)h->ntotal * h->stride);
+  uint8_t *dst = h->st_codes.as<uint8_t>() + (size_t)h->ntotal * h->stride;
+  if (h->type == T_IVFFLAT) {
+    hipLaunchKernelGGL(k_pack_rows, grid1d(n * h->d), dim3(256), 0, stream, x,
+                       n, h->d, h->stride, dst);
+  } else {
+    DevBuf resid;
+    resid.ensure((size_t)n * h->d * 4);
+    hipLaunchKernelGGL(k_residual, grid1d(n * h->d), dim3(256), 0, stream, x,
+                       h->centroids.as<float>(), asg.as<int>(), n, h->d,
+                       resid.as<float>());
+    if (h->type == T_IVFPQ) {
+      size_t lds = (size_t)256 * h->dsub * 4 <= 48 * 1024
+                       ? (size_t)256 * h->dsub * 4 : 0;
+      hipLaunchKernelGGL(k_pq_encode, grid1d(n), dim3(256), lds, stream,
+                         resid.as<float>(), h->codebooks.as<float>(), n, h->d,
+                         h->m, h->dsub, h->stride, dst);
+    } else {
+      hipLaunchKernelGGL(k_sq_encode, grid1d(n * h->d), dim3(256), 0, stream,
+                         resid.as<float>(),
+                         h->sq8 ? h->sq_vmin.as<float>() : nullptr,
+                         h->sq8 ? h->sq_vdiff.as<float>() : nullptr, n, h->d,
+                         h->stride, h->sq8 ? 0 : 1, dst);
+    }
+  }
+  HIP_CHECK(hipGetLastError());
+  // download assignments (arrival-order host copy drives finalize)
+  size_t old = h->h_assign.size();
+  h->h_assign.resize(old + n);
+  HIP_CHECK(hipMemcpyAsync(h->h_assign.data() + old, asg.p, (size_t)n * 4,
+                           hipMemcpyDeviceToHost, stream));
+  HIP_CHECK(hipStreamSynchronize(stream));
+  h->ntotal += n;
+  h->dirty = true;
+}
+
+// rebuild CSR from staging (stable counting sort by list)
+static void finalize_csr(dfann_index *h, hipStream_t stream) {
+  if (!h->dirty) return;
+  int64_t n = h->ntotal;
+  h->h_off.assign(h->nlist + 1, 0);
+  for (int64_t i = 0; i < n; ++i) h->h_off[h->h_assign[i] + 1]++;
+  for (int l = 0; l < h->nlist; ++l) h->h_off[l + 1] += h->h_off[l];
+  std::vector<unsigned> src(n);
+  {
+    std::vector<int64_t> fill(h->h_off.begin(), h->h_off.end() - 1);
+    for (int64_t i = 0; i < n; ++i) src[fill[h->h_assign[i]]++] = (unsigned)i;
+  }
+  h->cr_codes.ensure((size_t)n * h->stride);
+  h->cr_ids.ensure((size_t)n * 8);
+  h->id2pos.ensure((size_t)n * 4);
+  h->cr_off.ensure((size_t)(h->nlist + 1) * 8);
+  DevBuf src_dev;
+  src_dev.ensure((size_t)n * 4);
+  HIP_CHECK(hipMemcpyAsync(src_dev.p, src.data(), (size_t)n * 4,
+                           hipMemcpyHostToDevice, stream));
+  HIP_CHECK(hipMemcpyAsync(h->cr_off.p, h->h_off.data(),
+                           (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice,
+                           stream));
+  hipLaunchKernelGGL(k_gather_finalize, grid1d(n), dim3(256), 0, stream,
+                     h->st_codes.as<uint8_t>(), src_dev.as<unsigned>(), n,
+                     h->stride, h->cr_codes.as<uint8_t>(),
+                     h->cr_ids.as<int64_t>(), h->id2pos.as<unsigned>());
+  HIP_CHECK(hipStreamSynchronize(stream));
+  h->dirty = false;
+}
+
+// ---------------------------------------------------------------------------
+// search
+// ---------------------------------------------------------------------------
+
+static void pad_fill(dfann_index *h, int64_t nq, int k, float *D, int64_t *I,
+                     hipStream_t stream) {
+  // empty index: all pads — run the merge kernel on zero candidates
+  hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256), SEL_LDS_BYTES,
+                     stream, (const float *)nullptr, (const unsigned *)nullptr,
+                     nq, 0, k, (const int64_t *)nullptr,
+                     h->metric == M_IP ? 1 : 0, D, I);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void coarse_impl(dfann_index *h, int64_t nq, const float *q, int nprobe,
+                        int32_t *probes, float *keys, hipStream_t stream) {
+  int nlist = h->nlist;
+  int64_t chunk = std::max<int64_t>(1, (512LL << 20) / ((int64_t)nlist * 4));
+  chunk = std::min<int64_t>(chunk, nq);
+  h->ws1.ensure((size_t)chunk * nlist * 4);
+  float *sc = h->ws1.as<float>();
+  for (int64_t s = 0; s < nq; s += chunk) {
+    int64_t c = std::min(chunk, nq - s);
+    gemm_keys(h, q + s * h->d, c, h->centroids.as<float>(), nlist, h->d,
+              h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, sc,
+              stream);
+    hipLaunchKernelGGL(k_topk_rows, dim3((unsigned)c), dim3(256), SEL_LDS_BYTES,
+                       stream, sc, c, (long long)nlist, (long long)nlist,
+                       nprobe, 0u, (long long)nprobe, keys + s * nprobe,
+                       reinterpret_cast<unsigned *>(probes) + s * nprobe);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
+                           int nprobe, const int32_t *probes,
+                           const float *keys, int k, float *D, int64_t *I,
+                           hipStream_t stream) {
+  // candidate arrays
+  h->ws3.ensure((size_t)nq * nprobe * k * 4);
+  h->ws4.ensure((size_t)nq * nprobe * k * 4);
+  float *cand_d = h->ws3.as<float>();
+  unsigned *cand_p = h->ws4.as<unsigned>();
+  int fam_floats;
+  void (*kern)(const float *, const float *, const float *, const float *,
+               const float *, const int *, const float *, const uint8_t *,
+               const int64_t *, int, int, int, int, int, int, int, float *,
+               unsigned *, int) = nullptr;
+  bool ip = h->metric == M_IP;
+  switch (h->type) {
+    case T_IVFPQ:
+      fam_floats = h->m * 256 + h->d;
+      kern = ip ? k_scan_pq_ip : k_scan_pq_l2;
+      break;
+    case T_IVFFLAT:
+      fam_floats = h->d;
+      kern = ip ? k_scan_ivfflat_ip : k_scan_ivfflat_l2;
+      break;
+    case T_IVFSQ:
+      if (h->sq8) {
+        fam_floats = 3 * h->d;
+        kern = ip ? k_scan_sq8_ip : k_scan_sq8_l2;
+      } else {
+        fam_floats = h->d;
+        kern = ip ? k_scan_sqf_ip : k_scan_sqf_l2;
+      }
+      break;
+    default:
+      throw std::runtime_error("scan on flat index");
+  }
+  size_t lds = (size_t)fam_floats * 4 + SEL_LDS_BYTES;
+  if (lds > 160 * 1024)
+    throw std::runtime_error("scan LDS over budget (m too large)");
+  TimingEv e;
+  if (h->timing) e = h->ev_begin(stream);
+  hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe)), dim3(256), lds,
+                     stream, q, h->centroids.as<float>(),
+                     h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                     h->sq_scale.as<float>(), probes, keys,
+                     h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
+                     (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride, cand_d,
+                     cand_p, fam_floats);
+  if (h->timing) {
+    h->ev_end(e, stream, h->ev_scan);
+    // algorithmic units: sum of probed list lengths
+    std::vector<int32_t> hp((size_t)nq * nprobe);
+    HIP_CHECK(hipMemcpy(hp.data(), probes, hp.size() * 4, hipMemcpyDeviceToHost));
+    int64_t rows = 0;
+    for (int32_t L : hp) rows += h->h_off[L + 1] - h->h_off[L];
+    h->scan_rows += rows;
+    h->scan_bytes += rows * h->stride;
+  }
+  TimingEv em;
+  if (h->timing) em = h->ev_begin(stream);
+  hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256), SEL_LDS_BYTES,
+                     stream, cand_d, cand_p, nq, nprobe * k, k,
+                     h->cr_ids.as<int64_t>(), ip ? 1 : 0, D, I);
+  if (h->timing) h->ev_end(em, stream, h->ev_merge);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void flat_search_impl(dfann_index *h, int64_t nq, const float *q, int k,
+                             float *D, int64_t *I, hipStream_t stream) {
+  if (h->ntotal == 0) { pad_fill(h, nq, k, D, I, stream); return; }
+  bool ip = h->metric == M_IP;
+  const int64_t CH = 65536;
+  int64_t nch = (h->ntotal + CH - 1) / CH;
+  // chunk queries so scores fit
+  int64_t qch = std::max<int64_t>(1, (512LL << 20) / (CH * 4));
+  qch = std::min(qch, nq);
+  h->ws1.ensure((size_t)qch * CH * 4);
+  h->ws2.ensure((size_t)std::max<int64_t>(CH, nq) * 4);            // bnorm / qnorm
+  h->ws3.ensure((size_t)nq * nch * k * 4);                         // chunk winners d
+  h->ws4.ensure((size_t)nq * nch * k * 4);                         // chunk winners p
+  h->ws5.ensure((size_t)nq * 4);                                   // qnorm
+  float *sc = h->ws1.as<float>();
+  float *bn = h->ws2.as<float>();
+  float *qn = h->ws5.as<float>();
+  float *wd = h->ws3.as<float>();
+  unsigned *wp = h->ws4.as<unsigned>();
+  if (!ip) rownorms(q, nq, h->d, qn, stream);
+  for (int64_t ci = 0; ci < nch; ++ci) {
+    int64_t b0 = ci * CH;
+    int64_t bn_rows = std::min(CH, h->ntotal - b0);
+    const float *base = h->flat.as<float>() + (size_t)b0 * h->d;
+    if (!ip) rownorms(base, bn_rows, h->d, bn, stream);
+    for (int64_t s = 0; s < nq; s += qch) {
+      int64_t c = std::min(qch, nq - s);
+      gemm_keys(h, q + s * h->d, c, base, bn_rows, h->d, bn, qn + s,
+                ip ? 0 : 2, sc, stream);
+      // winners layout: (nq, nch, k) — row stride nch*k
+      hipLaunchKernelGGL(k_topk_rows, dim3((unsigned)c), dim3(256),
+                         SEL_LDS_BYTES, stream, sc, c, (long long)bn_rows,
+                         (long long)bn_rows, k, (unsigned)b0,
+                         (long long)(nch * k), wd + (s * nch + ci) * k,
+                         wp + (s * nch + ci) * k);
+    }
+  }
+  // merge chunk winners
+  hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256), SEL_LDS_BYTES,
+                     stream, wd, wp, nq, (int)(nch * k), k,
+                     (const int64_t *)nullptr, ip ? 1 : 0, D, I);
+  HIP_CHECK(hipGetLastError());
+}
+
+static void search_impl(dfann_index *h, int64_t nq, const float *q, int k,
+                        float *D, int64_t *I, hipStream_t stream) {
+  if (!h->trained) throw std::runtime_error("search on untrained index");
+  if (k > 512) throw std::runtime_error("k > 512 unsupported");
+  if (nq == 0) return;
+  if (h->type == T_FLAT) { flat_search_impl(h, nq, q, k, D, I, stream); return; }
+  if (h->ntotal == 0) { pad_fill(h, nq, k, D, I, stream); return; }
+  finalize_csr(h, stream);
+  int nprobe = std::min(h->nprobe, h->nlist);
+  if (nprobe > 512) nprobe = 512;
+  // probes + keys
+  DevBuf &pb = h->ws2;
+  pb.ensure((size_t)nq * nprobe * 8);
+  int32_t *probes = pb.as<int32_t>();
+  float *keys = reinterpret_cast<float *>(pb.as<uint8_t>() + (size_t)nq * nprobe * 4);
+  coarse_impl(h, nq, q, nprobe, probes, keys, stream);
+  scan_and_merge(h, nq, q, nprobe, probes, keys, k, D, I, stream);
+}
+
+// ---------------------------------------------------------------------------
+// C-ABI
+// ---------------------------------------------------------------------------
+
+#define API_BEGIN try {
+#define API_END                                                                \
+  return 0;                                                                    \
+  }                                                                            \
+  catch (const std::exception &e) { g_err = e.what(); return 1; }
+
+extern "C" int dfann_create(const char *spec_json, dfann_index **out) {
+  API_BEGIN
+  *out = create_from_spec(spec_json);
+  API_END
+}
+
+extern "C" int dfann_destroy(dfann_index *h) {
+  API_BEGIN
+  delete h;
+  API_END
+}
+
+extern "C" int dfann_train(dfann_index *h, int64_t n, const float *x_dev,
+                           dfann_stream stream) {
+  API_BEGIN
+  train_impl(h, n, x_dev, (hipStream_t)stream);
+  API_END
+}
+
+extern "C" int dfann_add(dfann_index *h, int64_t n, const float *x_dev,
+                         dfann_stream stream) {
+  API_BEGIN
+  add_impl(h, n, x_dev, (hipStream_t)stream);
+  API_END
+}
+
+extern "C" int dfann_search(dfann_index *h, int64_t nq, const float *q_dev,
+                            int k, float *D_dev, int64_t *I_dev,
+                            dfann_stream stream) {
+  API_BEGIN
+  search_impl(h, nq, q_dev, k, D_dev, I_dev, (hipStream_t)stream);
+  API_END
+}
+
+extern "C" int dfann_coarse(dfann_index *h, int64_t nq, const float *q_dev,
+                            int nprobe, int32_t *probes_dev, float *keys_dev,
+                            dfann_stream stream) {
+  API_BEGIN
+  if (!h->trained || h->type == T_FLAT)
+    throw std::runtime_error("coarse needs a trained IVF index");
+  nprobe = std::min(nprobe, h->nlist);
+  coarse_impl(h, nq, q_dev, nprobe, probes_dev, keys_dev, (hipStream_t)stream);
+  API_END
+}
+
+extern "C" int dfann_search_preassigned(dfann_index *h, int64_t nq,
+                                        const float *q_dev, int nprobe,
+                                        const int32_t *probes_dev,
+                                        const float *keys_dev, int k,
+                                        float *D_dev, int64_t *I_dev,
+                                        dfann_stream stream) {
+  API_BEGIN
+  if (!h->trained || h->type == T_FLAT)
+    throw std::runtime_error("search_preassigned needs a trained IVF index");
+  if (h->ntotal == 0) {
+    pad_fill(h, nq, k, D_dev, I_dev, (hipStream_t)stream);
+    return 0;
+  }
+  finalize_csr(h, (hipStream_t)stream);
+  scan_and_merge(h, nq, q_dev, nprobe, probes_dev, keys_dev, k, D_dev, I_dev,
+                 (hipStream_t)stream);
+  API_END
+}
+
+extern "C" int dfann_search_reconstruct(dfann_index *h, int64_t nq,
+                                        const float *q_dev, int k,
+                                        float *D_dev, int64_t *I_dev,
+                                        float *R_dev, dfann_stream stream) {
+  API_BEGIN
+  search_impl(h, nq, q_dev, k, D_dev, I_dev, (hipStream_t)stream);
+  int rtype;
+  const float *flat_src = nullptr;
+  if (h->type == T_FLAT) { rtype = 0; flat_src = h->flat.as<float>(); }
+  else if (h->type == T_IVFFLAT) rtype = 0;
+  else if (h->type == T_IVFPQ) rtype = 2;
+  else rtype = h->sq8 ? 3 : 4;
+  hipLaunchKernelGGL(k_reconstruct, dim3((unsigned)(nq * k)), dim3(64), 0,
+                     (hipStream_t)stream, I_dev, nq, k, rtype, h->d, h->m,
+                     h->dsub, h->stride, flat_src, h->cr_codes.as<uint8_t>(),
+                     h->id2pos.as<unsigned>(), h->cr_off.as<int64_t>(),
+                     h->nlist, h->centroids.as<float>(),
+                     h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                     h->sq_scale.as<float>(), R_dev);
+  HIP_CHECK(hipGetLastError());
+  API_END
+}
+
+extern "C" int dfann_set_nprobe(dfann_index *h, int nprobe) {
+  API_BEGIN
+  h->nprobe = nprobe;
+  API_END
+}
+
+extern "C" int64_t dfann_ntotal(dfann_index *h) { return h->ntotal; }
+extern "C" int dfann_nlist(dfann_index *h) { return h->nlist; }
+extern "C" int dfann_is_trained(dfann_index *h) { return h->trained ? 1 : 0; }
+extern "C" int dfann_dim(dfann_index *h) { return h->d; }
+extern "C" const char *dfann_spec_json(dfann_index *h) {
+  return h->spec_json.c_str();
+}
+
+extern "C" int dfann_get_centroids(dfann_index *h, float *out_host) {
+  API_BEGIN
+  if (h->type == T_FLAT)
+    throw std::runtime_error("'flat' index has no quantizer");
+  if (!h->trained) throw std::runtime_error("index not trained");
+  HIP_CHECK(hipMemcpy(out_host, h->centroids.p, (size_t)h->nlist * h->d * 4,
+                      hipMemcpyDeviceToHost));
+  API_END
+}
+
+extern "C" int dfann_get_codebooks(dfann_index *h, float *out_host) {
+  API_BEGIN
+  if (h->type != T_IVFPQ) throw std::runtime_error("not an ivfpq index");
+  HIP_CHECK(hipMemcpy(out_host, h->codebooks.p,
+                      (size_t)h->m * 256 * h->dsub * 4, hipMemcpyDeviceToHost));
+  API_END
+}
+
+extern "C" int dfann_get_sq_params(dfann_index *h, float *vmin_host,
+                                   float *vdiff_host) {
+  API_BEGIN
+  if (!(h->type == T_IVFSQ && h->sq8))
+    throw std::runtime_error("not an 8-bit ivfsq index");
+  HIP_CHECK(hipMemcpy(vmin_host, h->sq_vmin.p, (size_t)h->d * 4,
+                      hipMemcpyDeviceToHost));
+  HIP_CHECK(hipMemcpy(vdiff_host, h->sq_vdiff.p, (size_t)h->d * 4,
+                      hipMemcpyDeviceToHost));
+  API_END
+}
+
+extern "C" int dfann_set_trained(dfann_index *h, const float *centroids_host,
+                                 const float *codebooks_host,
+                                 const float *vmin_host,
+                                 const float *vdiff_host) {
+  API_BEGIN
+  if (h->type == T_FLAT) { h->trained = true; return 0; }
+  if (!centroids_host) throw std::runtime_error("centroids required");
+  h->centroids.ensure((size_t)h->nlist * h->d * 4);
+  h->cnorm.ensure((size_t)h->nlist * 4);
+  HIP_CHECK(hipMemcpy(h->centroids.p, centroids_host,
+                      (size_t)h->nlist * h->d * 4, hipMemcpyHostToDevice));
+  rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
+  if (h->type == T_IVFPQ) {
+    if (!codebooks_host) throw std::runtime_error("codebooks required");
+    h->codebooks.ensure((size_t)h->m * 256 * h->dsub * 4);
+    HIP_CHECK(hipMemcpy(h->codebooks.p, codebooks_host,
+                        (size_t)h->m * 256 * h->dsub * 4,
+                        hipMemcpyHostToDevice));
+  }
+  if (h->type == T_IVFSQ && h->sq8) {
+    if (!vmin_host || !vdiff_host) throw std::runtime_error("sq params required");
+    h->sq_vmin.ensure((size_t)h->d * 4);
+    h->sq_vdiff.ensure((size_t)h->d * 4);
+    h->sq_scale.ensure((size_t)h->d * 4);
+    HIP_CHECK(hipMemcpy(h->sq_vmin.p, vmin_host, (size_t)h->d * 4,
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpy(h->sq_vdiff.p, vdiff_host, (size_t)h->d * 4,
+                        hipMemcpyHostToDevice));
+    std::vector<float> sc(h->d);
+    std::vector<float> vd(h->d);
+    memcpy(vd.data(), vdiff_host, (size_t)h->d * 4);
+    for (int t = 0; t < h->d; ++t) sc[t] = vd[t] / 255.0f;
+    HIP_CHECK(hipMemcpy(h->sq_scale.p, sc.data(), (size_t)h->d * 4,
+                        hipMemcpyHostToDevice));
+  }
+  HIP_CHECK(hipDeviceSynchronize());
+  h->trained = true;
+  API_END
+}
+
+// ---------------------------------------------------------------------------
+// shard merge
+// ---------------------------------------------------------------------------
+
+extern "C" int dfann_merge_topk(int64_t nq, int S, int k, const float *D_dev,
+                                const int64_t *I_dev, int maximize,
+                                float *Dout_dev, int64_t *Iout_dev,
+                                dfann_stream stream) {
+  API_BEGIN
+  (void)I_dev;  // slots map to metadata host-side (ref client.py:297-298)
+  if (k > 512) throw std::runtime_error("k > 512 unsupported");
+  hipLaunchKernelGGL(k_merge_shards, dim3((unsigned)nq), dim3(256),
+                     SEL_LDS_BYTES, (hipStream_t)stream, D_dev, nq, S, k,
+                     maximize, Dout_dev, Iout_dev);
+  HIP_CHECK(hipGetLastError());
+  API_END
+}
+
+// ---------------------------------------------------------------------------
+// persistence (our own format, version 1)
+// ---------------------------------------------------------------------------
+
+static void fwrite_chk(const void *p, size_t n, FILE *f) {
+  if (fwrite(p, 1, n, f) != n) throw std::runtime_error("short write");
+}
+static void fread_chk(void *p, size_t n, FILE *f) {
+  if (fread(p, 1, n, f) != n) throw std::runtime_error("short read");
+}
+
+static void dump_dev(FILE *f, DevBuf &b, size_t bytes) {
+  std::vector<char> tmp(bytes);
+  if (bytes) HIP_CHECK(hipMemcpy(tmp.data(), b.p, bytes, hipMemcpyDeviceToHost));
+  fwrite_chk(tmp.data(), bytes, f);
+}
+static void load_dev(FILE *f, DevBuf &b, size_t bytes) {
+  std::vector<char> tmp(bytes);
+  fread_chk(tmp.data(), bytes, f);
+  b.ensure(std::max(bytes, (size_t)16));
+  if (bytes) HIP_CHECK(hipMemcpy(b.p, tmp.data(), bytes, hipMemcpyHostToDevice));
+}
+
+extern "C" int dfann_save(dfann_index *h, const char *path) {
+  API_BEGIN
+  finalize_csr(h, 0);
+  FILE *f = fopen(path, "wb");
+  if (!f) throw std::runtime_error(std::string("cannot open ") + path);
+  try {
+    const uint64_t magic = 0x31304E4E414644ULL;  // "DFANN01"
+    fwrite_chk(&magic, 8, f);
+    uint64_t jlen = h->spec_json.size();
+    fwrite_chk(&jlen, 8, f);
+    fwrite_chk(h->spec_json.data(), jlen, f);
+    int32_t hdr[8] = {h->type, h->metric, h->d, h->nlist, h->m,
+                      h->nprobe, h->sq8 ? 1 : 0, h->stride};
+    fwrite_chk(hdr, sizeof(hdr), f);
+    uint8_t tr = h->trained ? 1 : 0;
+    fwrite_chk(&tr, 1, f);
+    fwrite_chk(&h->ntotal, 8, f);
+    if (h->trained && h->type != T_FLAT) {
+      dump_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
+      if (h->type == T_IVFPQ)
+        dump_dev(f, h->codebooks, (size_t)h->m * 256 * h->dsub * 4);
+      if (h->type == T_IVFSQ && h->sq8) {
+        dump_dev(f, h->sq_vmin, (size_t)h->d * 4);
+        dump_dev(f, h->sq_vdiff, (size_t)h->d * 4);
+      }
+    }
+    if (h->type == T_FLAT) {
+      dump_dev(f, h->flat, (size_t)h->ntotal * h->d * 4);
+    } else if (h->ntotal) {
+      fwrite_chk(h->h_off.data(), (size_t)(h->nlist + 1) * 8, f);
+      dump_dev(f, h->cr_ids, (size_t)h->ntotal * 8);
+      dump_dev(f, h->cr_codes, (size_t)h->ntotal * h->stride);
+    }
+  } catch (...) {
+    fclose(f);
+    throw;
+  }
+  fclose(f);
+  API_END
+}
+
+extern "C" int dfann_load(const char *path, dfann_index **out) {
+  API_BEGIN
+  FILE *f = fopen(path, "rb");
+  if (!f) throw std::runtime_error(std::string("cannot open ") + path);
+  dfann_index *h = nullptr;
+  try {
+    uint64_t magic;
+    fread_chk(&magic, 8, f);
+    if (magic != 0x31304E4E414644ULL)
+      throw std::runtime_error("bad dfann file magic");
+    uint64_t jlen;
+    fread_chk(&jlen, 8, f);
+    std::string js(jlen, 0);
+    fread_chk(&js[0], jlen, f);
+    h = create_from_spec(js);
+    int32_t hdr[8];
+    fread_chk(hdr, sizeof(hdr), f);
+    h->nprobe = hdr[5];
+    uint8_t tr;
+    fread_chk(&tr, 1, f);
+    fread_chk(&h->ntotal, 8, f);
+    h->trained = tr != 0;
+    if (h->trained && h->type != T_FLAT) {
+      load_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
+      h->cnorm.ensure((size_t)h->nlist * 4);
+      rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
+      if (h->type == T_IVFPQ)
+        load_dev(f, h->codebooks, (size_t)h->m * 256 * h->dsub * 4);
+      if (h->type == T_IVFSQ && h->sq8) {
+        load_dev(f, h->sq_vmin, (size_t)h->d * 4);
+        load_dev(f, h->sq_vdiff, (size_t)h->d * 4);
+        std::vector<float> vd(h->d), sc(h->d);
+        HIP_CHECK(hipMemcpy(vd.data(), h->sq_vdiff.p, (size_t)h->d * 4,
+                            hipMemcpyDeviceToHost));
+        for (int t = 0; t < h->d; ++t) sc[t] = vd[t] / 255.0f;
+        h->sq_scale.ensure((size_t)h->d * 4);
+        HIP_CHECK(hipMemcpy(h->sq_scale.p, sc.data(), (size_t)h->d * 4,
+                            hipMemcpyHostToDevice));
+      }
+    }
+    if (h->type == T_FLAT) {
+      load_dev(f, h->flat, (size_t)h->ntotal * h->d * 4);
+    } else if (h->ntotal) {
+      h->h_off.resize(h->nlist + 1);
+      fread_chk(h->h_off.data(), (size_t)(h->nlist + 1) * 8, f);
+      load_dev(f, h->cr_ids, (size_t)h->ntotal * 8);
+      load_dev(f, h->cr_codes, (size_t)h->ntotal * h->stride);
+      h->cr_off.ensure((size_t)(h->nlist + 1) * 8);
+      HIP_CHECK(hipMemcpy(h->cr_off.p, h->h_off.data(),
+                          (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice));
+      h->id2pos.ensure((size_t)h->ntotal * 4);
+      // rebuild arrival-order staging + h_assign from CSR
+      h->st_codes.ensure((size_t)h->ntotal * h->stride);
+      hipLaunchKernelGGL(k_scatter_rows, grid1d(h->ntotal), dim3(256), 0, 0,
+                         h->cr_codes.as<uint8_t>(), h->cr_ids.as<int64_t>(),
+                         h->ntotal, h->stride, h->st_codes.as<uint8_t>());
+      std::vector<int64_t> ids(h->ntotal);
+      HIP_CHECK(hipMemcpy(ids.data(), h->cr_ids.p, (size_t)h->ntotal * 8,
+                          hipMemcpyDeviceToHost));
+      h->h_assign.resize(h->ntotal);
+      std::vector<unsigned> i2p(h->ntotal);
+      for (int l = 0; l < h->nlist; ++l)
+        for (int64_t j = h->h_off[l]; j < h->h_off[l + 1]; ++j) {
+          h->h_assign[ids[j]] = l;
+          i2p[ids[j]] = (unsigned)j;
+        }
+      HIP_CHECK(hipMemcpy(h->id2pos.p, i2p.data(), (size_t)h->ntotal * 4,
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipDeviceSynchronize());
+    }
+  } catch (...) {
+    fclose(f);
+    delete h;
+    throw;
+  }
+  fclose(f);
+  *out = h;
+  API_END
+}
+
+// ---------------------------------------------------------------------------
+// timing
+// ---------------------------------------------------------------------------
+
+extern "C" int dfann_set_timing(dfann_index *h, int enabled) {
+  API_BEGIN
+  h->timing = enabled != 0;
+  API_END
+}
+
+static double sum_events(std::vector<TimingEv> &v) {
+  double ms = 0;
+  for (auto &e : v) {
+    HIP_CHECK(hipEventSynchronize(e.b));
+    float el = 0;
+    HIP_CHECK(hipEventElapsedTime(&el, e.a, e.b));
+    ms += el;
+    hipEventDestroy(e.a);
+    hipEventDestroy(e.b);
+  }
+  v.clear();
+  return ms;
+}
+
+extern "C" int dfann_get_timing(dfann_index *h, dfann_timing *out) {
+  API_BEGIN
+  out->scan_launches = (int64_t)h->ev_scan.size();
+  out->merge_launches = (int64_t)h->ev_merge.size();
+  out->scan_ms = sum_events(h->ev_scan);
+  out->gemm_ms = sum_events(h->ev_gemm);
+  out->merge_ms = sum_events(h->ev_merge);
+  out->scan_rows = h->scan_rows;
+  out->scan_bytes = h->scan_bytes;
+  out->gemm_flops = h->gemm_flops;
+  h->scan_rows = h->scan_bytes = h->gemm_flops = 0;
+  API_END
+}

@@ -1,0 +1,894 @@
+// dfann — gfx950 (MI355X/CDNA4) kernels for the sharded ANN hot path.
+//
+// Built from scratch for CDNA4: wave64 everywhere, MFMA f32
+// (v_mfma_f32_32x32x2_f32) for the coarse-quantizer / flat distance GEMM,
+// inverted-list scans with PQ LUTs / SQ codecs staged in LDS over packed
+// HBM code reads, and a threshold-filtered LDS candidate buffer with
+// block-wide bitonic selection for every top-k stage.
+//
+// Numeric contract with the CPU oracle (oracle/core.py header): list-scan
+// accumulations are sequential over the reduced axis with fp contraction
+// off, so distances are bitwise equal to the oracle given shared trained
+// artifacts (PQ ADC, SQ8, SQfp16 paths). The IVF-Flat scan uses a
+// 16-lane-tree reduction (documented tolerance path).
+//
+// Tie-break everywhere: (distance, ascending id); minimize-keys internally
+// (IP negated), faiss sign conventions restored at the output edge.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <stdint.h>
+
+#define DFANN_BLOCK 256
+#define SEL_CAP 1024            // candidate buffer entries (k <= 512)
+#define PAD_POS 0xFFFFFFFFu
+#define DFANN_FLT_MAX 3.402823466e+38f
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// ---------------------------------------------------------------------------
+// selection machinery: threshold-filtered append + block bitonic compact
+// LDS carve (16B aligned): float selD[SEL_CAP]; unsigned selP[SEL_CAP];
+// int ctrl[4] = {cnt, pad, thrP, 0}; float thrD in ctrl-adjacent slot.
+// ---------------------------------------------------------------------------
+
+struct Sel {
+  float *d;       // SEL_CAP
+  unsigned *p;    // SEL_CAP
+  int *cnt;
+  float *thrD;
+  unsigned *thrP;
+};
+
+__device__ __forceinline__ Sel sel_carve(char *base) {
+  Sel s;
+  s.d = reinterpret_cast<float *>(base);
+  s.p = reinterpret_cast<unsigned *>(base + SEL_CAP * 4);
+  s.cnt = reinterpret_cast<int *>(base + SEL_CAP * 8);
+  s.thrD = reinterpret_cast<float *>(base + SEL_CAP * 8 + 4);
+  s.thrP = reinterpret_cast<unsigned *>(base + SEL_CAP * 8 + 8);
+  return s;
+}
+#define SEL_LDS_BYTES (SEL_CAP * 8 + 16)
+
+__device__ __forceinline__ void sel_init(Sel s) {
+  if (threadIdx.x == 0) {
+    *s.cnt = 0;
+    *s.thrD = DFANN_FLT_MAX;
+    *s.thrP = PAD_POS;
+  }
+  for (int i = threadIdx.x; i < SEL_CAP; i += blockDim.x) {
+    s.d[i] = DFANN_FLT_MAX;
+    s.p[i] = PAD_POS;
+  }
+}
+
+// lexicographic (dist, pos) — pads (FLT_MAX, PAD_POS) sort last
+__device__ __forceinline__ bool sel_less(float d0, unsigned p0, float d1, unsigned p1) {
+  return d0 < d1 || (d0 == d1 && p0 < p1);
+}
+
+__device__ __forceinline__ void sel_try(Sel s, float dist, unsigned pos) {
+  // caller guarantees capacity headroom; threshold is stable between compacts
+  if (sel_less(dist, pos, *s.thrD, *s.thrP)) {
+    int slot = atomicAdd(s.cnt, 1);
+    s.d[slot] = dist;
+    s.p[slot] = pos;
+  }
+}
+
+// block-wide: pad [cnt, CAP), bitonic sort CAP entries, truncate to k.
+// Requires a preceding __syncthreads() by the caller.
+__device__ void sel_compact(Sel s, int k) {
+  int cnt = *s.cnt;
+  __syncthreads();
+  for (int i = threadIdx.x; i < SEL_CAP; i += blockDim.x) {
+    if (i >= cnt) {
+      s.d[i] = DFANN_FLT_MAX;
+      s.p[i] = PAD_POS;
+    }
+  }
+  for (int kk = 2; kk <= SEL_CAP; kk <<= 1) {
+    for (int j = kk >> 1; j > 0; j >>= 1) {
+      __syncthreads();
+      for (int i = threadIdx.x; i < SEL_CAP; i += blockDim.x) {
+        int ixj = i ^ j;
+        if (ixj > i) {
+          bool asc = ((i & kk) == 0);
+          float di = s.d[i], dj = s.d[ixj];
+          unsigned pi = s.p[i], pj = s.p[ixj];
+          bool swap_needed = asc ? sel_less(dj, pj, di, pi) : sel_less(di, pi, dj, pj);
+          if (swap_needed) {
+            s.d[i] = dj; s.d[ixj] = di;
+            s.p[i] = pj; s.p[ixj] = pi;
+          }
+        }
+      }
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int newcnt = cnt < k ? cnt : k;
+    *s.cnt = newcnt;
+    if (newcnt >= k) {
+      *s.thrD = s.d[k - 1];
+      *s.thrP = s.p[k - 1];
+    }
+  }
+  __syncthreads();
+}
+
+// returns true if a compact happened (callers re-sync on their own)
+__device__ __forceinline__ void sel_guard(Sel s, int k, int max_appends) {
+  __syncthreads();
+  if (*s.cnt > SEL_CAP - max_appends) sel_compact(s, k);
+}
+
+// ---------------------------------------------------------------------------
+// GEMM (fp32 MFMA): C[i][j] = sum_k A[i][k] * B[j][k]   (A: MxK, B: NxK)
+// 128x128 tile, 4 waves x (2x2 of 32x32) on v_mfma_f32_32x32x2_f32.
+// Operand map (cdna_hip_programming.md §3): lane l feeds A[i=l&31][k=l>>5],
+// B[k=l>>5][j=l&31]; C/D: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5).
+// ---------------------------------------------------------------------------
+
+#define GT 128
+#define GK 32
+
+extern "C" __global__ __launch_bounds__(256) void k_gemm_nt(
+    const float *__restrict__ A, const float *__restrict__ B,
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc) {
+  __shared__ float sA[GT][GK + 1];
+  __shared__ float sB[GT][GK + 1];
+  int bi = blockIdx.y * GT;
+  int bj = blockIdx.x * GT;
+  int tid = threadIdx.x;
+  int lane = tid & 63, w = tid >> 6;
+  int wr = (w >> 1) * 64, wc = (w & 1) * 64;
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+  int rl = lane & 31;
+  int klane = lane >> 5;
+  for (int k0 = 0; k0 < K; k0 += GK) {
+    for (int e = tid; e < GT * GK; e += 256) {
+      int r = e >> 5, c = e & 31;  // GK == 32
+      sA[r][c] = (bi + r < M && k0 + c < K) ? A[(size_t)(bi + r) * lda + k0 + c] : 0.f;
+      sB[r][c] = (bj + r < N && k0 + c < K) ? B[(size_t)(bj + r) * ldb + k0 + c] : 0.f;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < GK; kk += 2) {
+      float a0 = sA[wr + rl][kk + klane];
+      float a1 = sA[wr + 32 + rl][kk + klane];
+      float b0 = sB[wc + rl][kk + klane];
+      float b1 = sB[wc + 32 + rl][kk + klane];
+      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  int row_in_tile_base = (lane >> 5) * 4;
+  int col_in_tile = lane & 31;
+#pragma unroll
+  for (int rg = 0; rg < 16; ++rg) {
+    int rit = (rg & 3) + 8 * (rg >> 2) + row_in_tile_base;
+    {
+      int row = bi + wr + rit, col = bj + wc + col_in_tile;
+      if (row < M && col < N) C[(size_t)row * ldc + col] = acc00[rg];
+      col = bj + wc + 32 + col_in_tile;
+      if (row < M && col < N) C[(size_t)row * ldc + col] = acc01[rg];
+      row = bi + wr + 32 + rit;
+      col = bj + wc + col_in_tile;
+      if (row < M && col < N) C[(size_t)row * ldc + col] = acc10[rg];
+      col = bj + wc + 32 + col_in_tile;
+      if (row < M && col < N) C[(size_t)row * ldc + col] = acc11[rg];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// row norms ||x_i||^2, one block per row
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(256) void k_rownorm(
+    const float *__restrict__ x, long long n, int d, float *__restrict__ out) {
+  long long row = blockIdx.x;
+  if (row >= n) return;
+  const float *xp = x + row * d;
+  float acc = 0.f;
+  for (int t = threadIdx.x; t < d; t += blockDim.x) {
+    float v = xp[t];
+    acc += v * v;
+  }
+#pragma unroll
+  for (int o = 32; o > 0; o >>= 1) acc += __shfl_down(acc, o, 64);
+  __shared__ float ws[4];
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  if (lane == 0) ws[w] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) out[row] = ws[0] + ws[1] + ws[2] + ws[3];
+}
+
+// ---------------------------------------------------------------------------
+// score epilogue: minimize-keys from raw inner products
+// mode 0: key = -ip (IP);  1: key = bn[j] - 2*ip (L2 rank);
+// mode 2: key = (qn[i] - 2*ip) + bn[j] (full squared L2, oracle op order)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void k_score_key(float *__restrict__ C, long long rows,
+                                       long long cols, long long ldc,
+                                       const float *__restrict__ qn,
+                                       const float *__restrict__ bn, int mode) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = rows * cols;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / cols, c = i % cols;
+    float v = C[r * ldc + c];
+    float key;
+    if (mode == 0) key = -v;
+    else if (mode == 1) key = bn[c] - 2.0f * v;
+    else key = (qn[r] - 2.0f * v) + bn[c];
+    C[r * ldc + c] = key;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// top-k per row over a key matrix (stream-select), pos = base + col.
+// out_d/out_p: (rows, k). grid.x = rows. dynamic LDS: SEL_LDS_BYTES.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(256) void k_topk_rows(
+    const float *__restrict__ keys, long long rows, long long cols,
+    long long ldk, int k, unsigned base, long long ldo,
+    float *__restrict__ out_d, unsigned *__restrict__ out_p) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  Sel s = sel_carve(smem);
+  long long row = blockIdx.x;
+  if (row >= rows) return;
+  sel_init(s);
+  __syncthreads();
+  const float *kp = keys + row * ldk;
+  for (long long c0 = 0; c0 < cols; c0 += 512) {
+    sel_guard(s, k, 512);
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      long long c = c0 + u * 256 + threadIdx.x;
+      if (c < cols) sel_try(s, kp[c], (unsigned)c + base);
+    }
+  }
+  __syncthreads();
+  sel_compact(s, k);
+  int cnt = *s.cnt;
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    bool v = j < cnt;
+    out_d[row * ldo + j] = v ? s.d[j] : DFANN_FLT_MAX;
+    out_p[row * ldo + j] = v ? s.p[j] : PAD_POS;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// running argmin across key-matrix chunks (assignment). rows = points.
+// best_v/best_i persist across chunk calls (init by k_fill_assign_init).
+// Ties: lowest global column (chunks ascending, strict <).
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void k_assign_init(float *best_v, int *best_i, long long n) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) {
+    best_v[i] = DFANN_FLT_MAX;
+    best_i[i] = -1;
+  }
+}
+
+extern "C" __global__ void k_assign_chunk(const float *__restrict__ keys,
+                                          long long rows, long long cols,
+                                          long long ldk, int col_base,
+                                          float *__restrict__ best_v,
+                                          int *__restrict__ best_i) {
+  long long r = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (r >= rows) return;
+  const float *kp = keys + r * ldk;
+  float bv = best_v[r];
+  int bi = best_i[r];
+  for (long long c = 0; c < cols; ++c) {
+    float v = kp[c];
+    if (v < bv) { bv = v; bi = col_base + (int)c; }
+  }
+  best_v[r] = bv;
+  best_i[r] = bi;
+}
+
+// ---------------------------------------------------------------------------
+// IVF list scan. One block per (query, probe). FAM: 0=PQ, 1=IVF-Flat,
+// 2=SQ8, 3=SQfp16. IP: minimize-key = -(bias + sum).
+// LDS: [fam region (fam_floats)] [SEL].
+// cand_d/cand_p: (nq, nprobe, k); pos = CSR row (uint32).
+// keys_dev: coarse minimize-keys (IP bias = -key); null for L2.
+// ---------------------------------------------------------------------------
+
+template <int FAM, bool IS_IP>
+__device__ void ivf_scan_body(
+    const float *__restrict__ q, const float *__restrict__ cent,
+    const float *__restrict__ cb, const float *__restrict__ sq_vmin,
+    const float *__restrict__ sq_scale, const int *__restrict__ probes,
+    const float *__restrict__ keys, const uint8_t *__restrict__ codes,
+    const int64_t *__restrict__ off, int nq, int nprobe, int d, int m,
+    int dsub, int k, int stride, float *__restrict__ cand_d,
+    unsigned *__restrict__ cand_p, int fam_floats) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float *fam = reinterpret_cast<float *>(smem);
+  Sel s = sel_carve(smem + (size_t)fam_floats * 4);
+  long long blk = blockIdx.x;
+  int bq = (int)(blk / nprobe), bp = (int)(blk % nprobe);
+  int L = probes[(long long)bq * nprobe + bp];
+  long long out_base = ((long long)bq * nprobe + bp) * k;
+  long long s0 = off[L], s1 = off[L + 1];
+  if (s0 == s1) {
+    for (int j = threadIdx.x; j < k; j += blockDim.x) {
+      cand_d[out_base + j] = DFANN_FLT_MAX;
+      cand_p[out_base + j] = PAD_POS;
+    }
+    return;
+  }
+  const float *qp = q + (long long)bq * d;
+  float bias = 0.f;
+  if (IS_IP) bias = -keys[(long long)bq * nprobe + bp];  // q . centroid
+
+  // --- stage FAM region ---
+  if (FAM == 0) {
+    // rbuf (d floats) then LUT (m*256) — rbuf lives after the LUT
+    float *lut = fam;
+    float *rbuf = fam + (size_t)m * 256;
+    for (int t = threadIdx.x; t < d; t += blockDim.x)
+      rbuf[t] = IS_IP ? qp[t] : qp[t] - cent[(long long)L * d + t];
+    __syncthreads();
+    for (int e = threadIdx.x; e < m * 256; e += blockDim.x) {
+      int j = e >> 8, c = e & 255;
+      const float *cbe = cb + ((size_t)j * 256 + c) * dsub;
+      const float *rs = rbuf + j * dsub;
+      float acc = 0.f;
+      for (int t = 0; t < dsub; ++t) {
+#pragma clang fp contract(off)
+        if (IS_IP) {
+          acc = acc + rs[t] * cbe[t];
+        } else {
+          float diff = rs[t] - cbe[t];
+          acc = acc + diff * diff;
+        }
+      }
+      lut[e] = acc;
+    }
+  } else if (FAM == 1) {
+    for (int t = threadIdx.x; t < d; t += blockDim.x) fam[t] = qp[t];
+  } else if (FAM == 2) {
+    float *rbuf = fam, *vm = fam + d, *sc = fam + 2 * d;
+    for (int t = threadIdx.x; t < d; t += blockDim.x) {
+      rbuf[t] = IS_IP ? qp[t] : qp[t] - cent[(long long)L * d + t];
+      vm[t] = sq_vmin[t];
+      sc[t] = sq_scale[t];
+    }
+  } else {  // SQfp16
+    for (int t = threadIdx.x; t < d; t += blockDim.x)
+      fam[t] = IS_IP ? qp[t] : qp[t] - cent[(long long)L * d + t];
+  }
+  sel_init(s);
+  __syncthreads();
+
+  // --- scan ---
+  if (FAM == 1) {
+    // 16-lane-per-vector tree reduction (tolerance parity path)
+    int sub = threadIdx.x & 15, grp = threadIdx.x >> 4;  // 16 groups
+    for (long long base = s0; base < s1; base += 16 * 8) {
+      sel_guard(s, k, 128);
+      for (int u = 0; u < 8; ++u) {
+        long long pos = base + (long long)u * 16 + grp;
+        float dist = 0.f;
+        bool valid = pos < s1;
+        if (valid) {
+          const float *vp = reinterpret_cast<const float *>(codes + pos * (size_t)stride);
+          float part = 0.f;
+          for (int t = sub; t < d; t += 16) {
+#pragma clang fp contract(off)
+            if (IS_IP) part = part + fam[t] * vp[t];
+            else {
+              float diff = fam[t] - vp[t];
+              part = part + diff * diff;
+            }
+          }
+#pragma unroll
+          for (int o = 8; o > 0; o >>= 1) part += __shfl_xor(part, o, 16);
+          dist = IS_IP ? -part : part;
+        }
+        if (valid && sub == 0) sel_try(s, dist, (unsigned)pos);
+      }
+    }
+  } else {
+    const float *lut = fam;                       // FAM 0
+    const float *rbuf = fam;                      // FAM 2/3 target vector
+    const float *vm = fam + d, *sc = fam + 2 * d; // FAM 2
+    for (long long base = s0; base < s1; base += 512) {
+      sel_guard(s, k, 512);
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        long long pos = base + (long long)u * 256 + threadIdx.x;
+        if (pos < s1) {
+          const uint8_t *cp = codes + pos * (size_t)stride;
+          float acc = 0.f;
+          if (FAM == 0) {
+#pragma clang fp contract(off)
+            for (int g = 0; g < m; g += 16) {
+              uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
+              unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
+#pragma unroll
+              for (int b = 0; b < 16; ++b) {
+                if (g + b < m) {
+                  unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
+                  unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
+                  acc = acc + lut[(g + b) * 256 + c];
+                }
+              }
+            }
+          } else if (FAM == 2) {
+#pragma clang fp contract(off)
+            for (int g = 0; g < d; g += 16) {
+              uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
+              unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
+#pragma unroll
+              for (int b = 0; b < 16; ++b) {
+                if (g + b < d) {
+                  unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
+                  unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
+                  int t = g + b;
+                  float dec = vm[t] + ((float)c + 0.5f) * sc[t];
+                  if (IS_IP) acc = acc + rbuf[t] * dec;
+                  else {
+                    float diff = rbuf[t] - dec;
+                    acc = acc + diff * diff;
+                  }
+                }
+              }
+            }
+          } else {  // FAM 3: fp16 codes
+#pragma clang fp contract(off)
+            for (int g = 0; g < d; g += 8) {  // 8 halves = 16 bytes
+              uint4 wv = *reinterpret_cast<const uint4 *>(cp + (size_t)g * 2);
+              unsigned ww[1];
+#pragma unroll
+              for (int b = 0; b < 8; ++b) {
+                if (g + b < d) {
+                  unsigned word = (b < 2) ? wv.x : (b < 4) ? wv.y : (b < 6) ? wv.z : wv.w;
+                  unsigned h = (word >> ((b & 1) * 16)) & 0xFFFFu;
+                  __half hv = __ushort_as_half((unsigned short)h);
+                  float dec = __half2float(hv);
+                  int t = g + b;
+                  if (IS_IP) acc = acc + rbuf[t] * dec;
+                  else {
+                    float diff = rbuf[t] - dec;
+                    acc = acc + diff * diff;
+                  }
+                }
+              }
+              (void)ww;
+            }
+          }
+          float dist = IS_IP ? -(bias + acc) : acc;
+          sel_try(s, dist, (unsigned)pos);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  sel_compact(s, k);
+  int cnt = *s.cnt;
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    bool v = j < cnt;
+    cand_d[out_base + j] = v ? s.d[j] : DFANN_FLT_MAX;
+    cand_p[out_base + j] = v ? s.p[j] : PAD_POS;
+  }
+}
+
+#define INSTANTIATE_SCAN(NAME, FAM, IS_IP)                                     \
+  extern "C" __global__ __launch_bounds__(256) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *codes, const int64_t *off, int nq,   \
+      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
+      unsigned *cand_p, int fam_floats) {                                      \
+    ivf_scan_body<FAM, IS_IP>(q, cent, cb, sq_vmin, sq_scale, probes, keys,    \
+                              codes, off, nq, nprobe, d, m, dsub, k, stride,   \
+                              cand_d, cand_p, fam_floats);                     \
+  }
+
+INSTANTIATE_SCAN(k_scan_pq_l2, 0, false)
+INSTANTIATE_SCAN(k_scan_pq_ip, 0, true)
+INSTANTIATE_SCAN(k_scan_ivfflat_l2, 1, false)
+INSTANTIATE_SCAN(k_scan_ivfflat_ip, 1, true)
+INSTANTIATE_SCAN(k_scan_sq8_l2, 2, false)
+INSTANTIATE_SCAN(k_scan_sq8_ip, 2, true)
+INSTANTIATE_SCAN(k_scan_sqf_l2, 3, false)
+INSTANTIATE_SCAN(k_scan_sqf_ip, 3, true)
+
+// ---------------------------------------------------------------------------
+// merge scan candidates -> final (D, I) per query.
+// cand entries: (minimize-key, pos); pos PAD_POS = skip.
+// ids: CSR ids array (pos -> arrival id); tie-break on arrival id.
+// If ids == nullptr, pos IS the id (flat chunk winners).
+// Output: faiss conventions (IP distances un-negated, pads).
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(256) void k_merge_cand(
+    const float *__restrict__ cand_d, const unsigned *__restrict__ cand_p,
+    long long nq, int C, int k, const int64_t *__restrict__ ids, int is_ip,
+    float *__restrict__ D, int64_t *__restrict__ I) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  Sel s = sel_carve(smem);
+  long long qi = blockIdx.x;
+  if (qi >= nq) return;
+  sel_init(s);
+  __syncthreads();
+  const float *cd = cand_d + qi * C;
+  const unsigned *cp = cand_p + qi * C;
+  for (int c0 = 0; c0 < C; c0 += 512) {
+    sel_guard(s, k, 512);
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int c = c0 + u * 256 + threadIdx.x;
+      if (c < C) {
+        unsigned pos = cp[c];
+        if (pos != PAD_POS) {
+          unsigned idu = ids ? (unsigned)ids[pos] : pos;
+          sel_try(s, cd[c], idu);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  sel_compact(s, k);
+  int cnt = *s.cnt;
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    bool v = j < cnt;
+    if (v) {
+      D[qi * k + j] = is_ip ? -s.d[j] : s.d[j];
+      I[qi * k + j] = (long long)s.p[j];
+    } else {
+      D[qi * k + j] = is_ip ? -DFANN_FLT_MAX : DFANN_FLT_MAX;
+      I[qi * k + j] = -1;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// shard-result merge (client-side heap semantics, ref client.py:265-310):
+// inputs (S, nq, k) faiss-convention D + ids; maximize negates on the way
+// in and the OUTPUT KEEPS the negation (reference quirk 2). Output ids =
+// slot s*nq*k + q*k + j (caller maps to shard metadata). Every input slot
+// is a candidate (pads carry FLT_MAX and can win, as in the reference).
+// Tie-break: ascending slot (= shard order, then rank).
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(256) void k_merge_shards(
+    const float *__restrict__ Dall, long long nq, int S, int k, int maximize,
+    float *__restrict__ Dout, int64_t *__restrict__ Iout) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  Sel s = sel_carve(smem);
+  long long qi = blockIdx.x;
+  if (qi >= nq) return;
+  sel_init(s);
+  __syncthreads();
+  int C = S * k;
+  for (int c0 = 0; c0 < C; c0 += 512) {
+    sel_guard(s, k, 512);
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int c = c0 + u * 256 + threadIdx.x;
+      if (c < C) {
+        int sh = c / k, j = c % k;
+        float v = Dall[((long long)sh * nq + qi) * k + j];
+        float key = maximize ? -v : v;
+        unsigned slot = (unsigned)(sh * k + j);  // dense per-query slot
+        sel_try(s, key, slot);
+      }
+    }
+  }
+  __syncthreads();
+  sel_compact(s, k);
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    unsigned slot = s.p[j];
+    int sh = slot / k, jj = slot % k;
+    Dout[qi * k + j] = s.d[j];  // negated for maximize — quirk 2 kept
+    Iout[qi * k + j] = ((long long)sh * nq + qi) * k + jj;  // global slot
+  }
+}
+
+// ---------------------------------------------------------------------------
+// encode / build kernels
+// ---------------------------------------------------------------------------
+
+// residual: out[i] = x[i] - cent[assign[i]]
+extern "C" __global__ void k_residual(const float *__restrict__ x,
+                                      const float *__restrict__ cent,
+                                      const int *__restrict__ assign,
+                                      long long n, int d,
+                                      float *__restrict__ out) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / d, t = i % d;
+    out[i] = x[i] - cent[(long long)assign[r] * d + t];
+  }
+}
+
+// PQ encode: one thread per point; codebook subspace staged in LDS when it
+// fits. resid: (n, d) residuals; codes out: (n, stride) u8.
+extern "C" __global__ __launch_bounds__(256) void k_pq_encode(
+    const float *__restrict__ resid, const float *__restrict__ cb, long long n,
+    int d, int m, int dsub, int stride, uint8_t *__restrict__ codes) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float *scb = reinterpret_cast<float *>(smem);  // 256*dsub when staged
+  bool stage = (size_t)256 * dsub * 4 <= 48 * 1024;
+  long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (int j = 0; j < m; ++j) {
+    const float *cbj = cb + (size_t)j * 256 * dsub;
+    if (stage) {
+      __syncthreads();
+      for (int e = threadIdx.x; e < 256 * dsub; e += blockDim.x) scb[e] = cbj[e];
+      __syncthreads();
+    }
+    const float *cbs = stage ? scb : cbj;
+    if (p < n) {
+      const float *rs = resid + p * d + j * dsub;
+      float best = DFANN_FLT_MAX;
+      int bc = 0;
+      for (int c = 0; c < 256; ++c) {
+        const float *ce = cbs + (size_t)c * dsub;
+        float acc = 0.f;
+        for (int t = 0; t < dsub; ++t) {
+          float diff = rs[t] - ce[t];
+          acc += diff * diff;
+        }
+        if (acc < best) { best = acc; bc = c; }
+      }
+      codes[p * (size_t)stride + j] = (uint8_t)bc;
+    }
+  }
+}
+
+// SQ encode (8bit / fp16) of residuals
+extern "C" __global__ void k_sq_encode(const float *__restrict__ resid,
+                                       const float *__restrict__ vmin,
+                                       const float *__restrict__ vdiff,
+                                       long long n, int d, int stride,
+                                       int is_fp16, uint8_t *__restrict__ codes) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / d;
+    int t = (int)(i % d);
+    float v = resid[i];
+    if (is_fp16) {
+      __half h = __float2half(v);
+      reinterpret_cast<unsigned short *>(codes + r * (size_t)stride)[t] =
+          __half_as_ushort(h);
+    } else {
+      float xi = (v - vmin[t]) / vdiff[t];
+      int c = (int)(255.0f * xi);  // trunc toward zero, as the oracle
+      c = c < 0 ? 0 : (c > 255 ? 255 : c);
+      codes[r * (size_t)stride + t] = (uint8_t)c;
+    }
+  }
+}
+
+// pack raw fp32 rows into the staging byte arena (IVF-Flat codes)
+extern "C" __global__ void k_pack_rows(const float *__restrict__ x, long long n,
+                                       int d, int stride,
+                                       uint8_t *__restrict__ out) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / d;
+    int t = (int)(i % d);
+    reinterpret_cast<float *>(out + r * (size_t)stride)[t] = x[i];
+  }
+}
+
+// finalize gather: CSR[j] = staging[src[j]]; ids[j] = src[j]; id2pos[src[j]] = j
+extern "C" __global__ void k_gather_finalize(const uint8_t *__restrict__ staging,
+                                             const unsigned *__restrict__ src,
+                                             long long n, int stride,
+                                             uint8_t *__restrict__ csr,
+                                             int64_t *__restrict__ ids,
+                                             unsigned *__restrict__ id2pos) {
+  long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= n) return;
+  unsigned sidx = src[j];
+  ids[j] = (long long)sidx;
+  id2pos[sidx] = (unsigned)j;
+  const uint4 *in = reinterpret_cast<const uint4 *>(staging + (size_t)sidx * stride);
+  uint4 *out = reinterpret_cast<uint4 *>(csr + (size_t)j * stride);
+  for (int t = 0; t < stride / 16; ++t) out[t] = in[t];
+}
+
+// inverse of finalize: rebuild the arrival-order staging arena from CSR
+extern "C" __global__ void k_scatter_rows(const uint8_t *__restrict__ csr,
+                                          const int64_t *__restrict__ ids,
+                                          long long n, int stride,
+                                          uint8_t *__restrict__ staging) {
+  long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= n) return;
+  const uint4 *in = reinterpret_cast<const uint4 *>(csr + (size_t)j * stride);
+  uint4 *out = reinterpret_cast<uint4 *>(staging + (size_t)ids[j] * stride);
+  for (int t = 0; t < stride / 16; ++t) out[t] = in[t];
+}
+
+// gather rows by index (f32): out[i] = in[idx[i]]
+extern "C" __global__ void k_gather_rows(const float *__restrict__ in,
+                                         const int64_t *__restrict__ idx,
+                                         long long n, int d,
+                                         float *__restrict__ out) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / d;
+    int t = (int)(i % d);
+    out[i] = in[idx[r] * d + t];
+  }
+}
+
+// strided subsample gather: out[i] = in[(i*n_in)/n_out]  (oracle kmeans)
+extern "C" __global__ void k_gather_strided(const float *__restrict__ in,
+                                            long long n_in, long long n_out,
+                                            int d, float *__restrict__ out) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n_out * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / d;
+    int t = (int)(i % d);
+    long long sr = (r * n_in) / n_out;
+    out[i] = in[sr * d + t];
+  }
+}
+
+// subspace slice: out[i*dsub + t] = in[i*d + j0 + t]
+extern "C" __global__ void k_subspace_slice(const float *__restrict__ in,
+                                            long long n, int d, int j0,
+                                            int dsub, float *__restrict__ out) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * dsub;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / dsub;
+    int t = (int)(i % dsub);
+    out[i] = in[r * d + j0 + t];
+  }
+}
+
+// k-means accumulation (fp32 atomics; order nondeterminism documented)
+extern "C" __global__ void k_centroid_accum(const float *__restrict__ x,
+                                            const int *__restrict__ assign,
+                                            long long n, int d,
+                                            float *__restrict__ sums,
+                                            int *__restrict__ counts) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long r = i / d;
+    int t = (int)(i % d);
+    atomicAdd(&sums[(long long)assign[r] * d + t], x[i]);
+    if (t == 0) atomicAdd(&counts[assign[r]], 1);
+  }
+}
+
+extern "C" __global__ void k_centroid_div(float *__restrict__ cent,
+                                          const float *__restrict__ sums,
+                                          const int *__restrict__ counts,
+                                          long long kcent, int d) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = kcent * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long c = i / d;
+    if (counts[c] > 0) cent[i] = sums[i] / (float)counts[c];
+  }
+}
+
+// per-dim min/max via monotone uint encoding (order-independent)
+__device__ __forceinline__ unsigned f32_enc(float f) {
+  unsigned u = __float_as_uint(f);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+__device__ __forceinline__ float f32_dec(unsigned u) {
+  unsigned v = (u & 0x80000000u) ? (u & 0x7FFFFFFFu) : ~u;
+  return __uint_as_float(v);
+}
+
+extern "C" __global__ void k_minmax_init(unsigned *mn, unsigned *mx, int d) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t < d) {
+    mn[t] = 0xFFFFFFFFu;
+    mx[t] = 0u;
+  }
+}
+
+extern "C" __global__ void k_minmax_dims(const float *__restrict__ x, long long n,
+                                         int d, unsigned *__restrict__ mn,
+                                         unsigned *__restrict__ mx) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = n * d;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    int t = (int)(i % d);
+    unsigned e = f32_enc(x[i]);
+    atomicMin(&mn[t], e);
+    atomicMax(&mx[t], e);
+  }
+}
+
+extern "C" __global__ void k_minmax_decode(const unsigned *mn, const unsigned *mx,
+                                           int d, float *vmin, float *vdiff,
+                                           float *scale) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t < d) {
+    float lo = f32_dec(mn[t]), hi = f32_dec(mx[t]);
+    float df = hi - lo;
+    if (df == 0.f) df = 1.0f;  // degenerate dim guard (oracle parity)
+    vmin[t] = lo;
+    vdiff[t] = df;
+    scale[t] = df / 255.0f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// reconstruction: decode result ids -> vectors. One block per (q, j).
+// type: 0 flat/ivfflat raw (flat_src != null uses flat arena directly by id,
+// else CSR codes row), 2 pq, 3 sq8, 4 sqfp16
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(64) void k_reconstruct(
+    const int64_t *__restrict__ I, long long nq, int k, int type, int d,
+    int m, int dsub, int stride, const float *__restrict__ flat_src,
+    const uint8_t *__restrict__ codes, const unsigned *__restrict__ id2pos,
+    const int64_t *__restrict__ off, int nlist,
+    const float *__restrict__ cent, const float *__restrict__ cb,
+    const float *__restrict__ vmin, const float *__restrict__ scale,
+    float *__restrict__ R) {
+  long long e = blockIdx.x;
+  if (e >= nq * k) return;
+  long long id = I[e];
+  float *out = R + e * d;
+  if (id < 0) {
+    for (int t = threadIdx.x; t < d; t += blockDim.x) out[t] = 0.f;
+    return;
+  }
+  if (flat_src) {
+    const float *src = flat_src + id * d;
+    for (int t = threadIdx.x; t < d; t += blockDim.x) out[t] = src[t];
+    return;
+  }
+  unsigned pos = id2pos[id];
+  // binary search list containing pos
+  int lo = 0, hi = nlist;
+  while (lo + 1 < hi) {
+    int mid = (lo + hi) >> 1;
+    if (off[mid] <= (long long)pos) lo = mid;
+    else hi = mid;
+  }
+  int L = lo;
+  const uint8_t *cp = codes + (size_t)pos * stride;
+  if (type == 0) {  // ivfflat raw
+    const float *src = reinterpret_cast<const float *>(cp);
+    for (int t = threadIdx.x; t < d; t += blockDim.x) out[t] = src[t];
+  } else if (type == 2) {  // pq: centroid + codebook
+    for (int t = threadIdx.x; t < d; t += blockDim.x) {
+      int j = t / dsub, tt = t % dsub;
+      unsigned c = cp[j];
+      out[t] = cent[(long long)L * d + t] + cb[((size_t)j * 256 + c) * dsub + tt];
+    }
+  } else if (type == 3) {  // sq8
+    for (int t = threadIdx.x; t < d; t += blockDim.x) {
+      unsigned c = cp[t];
+      out[t] = cent[(long long)L * d + t] + (vmin[t] + ((float)c + 0.5f) * scale[t]);
+    }
+  } else {  // sqfp16
+    const unsigned short *hp = reinterpret_cast<const unsigned short *>(cp);
+    for (int t = threadIdx.x; t < d; t += blockDim.x) {
+      out[t] = cent[(long long)L * d + t] + __half2float(__ushort_as_half(hp[t]));
+    }
+  }
+}
