@@ -847,6 +847,23 @@ extern "C" int dfann_get_codebooks(dfann_index *h, float *out_host) {
   API_END
 }
 
+extern "C" int dfann_get_lists(dfann_index *h, int64_t *off_host,
+                               int64_t *ids_host, uint8_t *codes_host) {
+  API_BEGIN
+  if (h->type == T_FLAT) throw std::runtime_error("flat index has no lists");
+  finalize_csr(h, 0);
+  memcpy(off_host, h->h_off.data(), (size_t)(h->nlist + 1) * 8);
+  if (h->ntotal) {
+    HIP_CHECK(hipMemcpy(ids_host, h->cr_ids.p, (size_t)h->ntotal * 8,
+                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(codes_host, h->cr_codes.p,
+                        (size_t)h->ntotal * h->stride, hipMemcpyDeviceToHost));
+  }
+  API_END
+}
+
+extern "C" int dfann_code_stride(dfann_index *h) { return h->stride; }
+
 extern "C" int dfann_get_sq_params(dfann_index *h, float *vmin_host,
                                    float *vdiff_host) {
   API_BEGIN

@@ -643,8 +643,9 @@ extern "C" __global__ __launch_bounds__(256) void k_pq_encode(
         const float *ce = cbs + (size_t)c * dsub;
         float acc = 0.f;
         for (int t = 0; t < dsub; ++t) {
+#pragma clang fp contract(off)
           float diff = rs[t] - ce[t];
-          acc += diff * diff;
+          acc = acc + diff * diff;
         }
         if (acc < best) { best = acc; bc = c; }
       }

@@ -1,0 +1,90 @@
+# Multi-GPU shard path: one process per GPU over torch.distributed
+# ("nccl" backend == RCCL over xGMI on MI355X).
+#
+# Replaces the reference's client-side fan-out + heap merge
+# (distributed_faiss/client.py:200-210,265-310) for the intra-node case:
+# every rank holds one shard (disjoint vector partition, matching the
+# reference's per-server partitioning, client.py:174-192), searches it
+# locally, then ONE exchange step: all-gather of per-shard (distance, id)
+# top-k (S*nq*k*12 B — latency-bound over xGMI) followed by
+# an on-GPU k-way merge (dfann_merge_topk) with the reference's
+# dot-negation convention (quirk 2).
+#
+# CPU path (gloo) exists for the world_size>1 correctness tests that run
+# without a GPU — the merge arithmetic there reuses the client's numpy
+# restatement, so the collective logic is identical.
+
+import os
+
+import numpy as np
+
+
+def init_from_env():
+    """Initialize torch.distributed from torchrun env vars; no-op when
+    WORLD_SIZE is absent or 1. Returns (rank, world_size)."""
+    import torch
+    import torch.distributed as dist
+
+    ws = int(os.environ.get("WORLD_SIZE", "1"))
+    if ws <= 1:
+        return 0, 1
+    if not dist.is_initialized():
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group(backend=backend)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+    return dist.get_rank(), ws
+
+
+def allgather_shard_topk(D_local, I_local):
+    """All-gather per-shard (nq,k) results -> (S,nq,k) tensors on every
+    rank. Tensors stay on device (RCCL over xGMI); gloo path on CPU."""
+    import torch
+    import torch.distributed as dist
+
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return D_local.unsqueeze(0), I_local.unsqueeze(0)
+    ws = dist.get_world_size()
+    Dg = [torch.empty_like(D_local) for _ in range(ws)]
+    Ig = [torch.empty_like(I_local) for _ in range(ws)]
+    dist.all_gather(Dg, D_local.contiguous())
+    dist.all_gather(Ig, I_local.contiguous())
+    return torch.stack(Dg), torch.stack(Ig)
+
+
+def merge_gathered(Dall, Iall, k, maximize):
+    """Merge (S,nq,k) shard results into (nq,k), reference heap semantics
+    (returned distances NEGATED for maximize — quirk 2). Returns
+    (D, shard_idx, local_ids): shard_idx/local_ids map each winner back to
+    its shard and per-shard id, mirroring client.py:290,297-298."""
+    if Dall.is_cuda:
+        from .hip_engine import merge_topk_dev
+
+        S, nq, kk = Dall.shape
+        Dm, slots = merge_topk_dev(Dall, Iall, k, maximize)
+        slots = slots.cpu().numpy()
+        # slot = s*nq*k + q*k + j
+        s_idx = slots // (nq * kk)
+        j_idx = slots % kk
+        q_idx = (slots // kk) % nq
+        Ia = Iall.cpu().numpy()
+        local = Ia[s_idx, q_idx, j_idx]
+        return Dm.cpu().numpy(), s_idx, local
+    # CPU (gloo tests): numpy restatement
+    Da = Dall.numpy()
+    Ia = Iall.numpy()
+    S, nq, kk = Da.shape
+    keys = -Da if maximize else Da
+    flat = keys.transpose(1, 0, 2).reshape(nq, S * kk)
+    slots = np.arange(S * kk)
+    Dout = np.empty((nq, k), dtype=np.float32)
+    s_out = np.empty((nq, k), dtype=np.int64)
+    l_out = np.empty((nq, k), dtype=np.int64)
+    for i in range(nq):
+        order = np.lexsort((slots, flat[i]))[:k]
+        Dout[i] = flat[i, order]
+        s_out[i] = order // kk
+        l_out[i] = Ia[order // kk, i, order % kk]
+    return Dout, s_out, l_out

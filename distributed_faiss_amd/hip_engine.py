@@ -167,6 +167,13 @@ class HipEngine:
             self._stream(torch)))
         torch.cuda.synchronize()
 
+    def train_dev(self, xt):
+        torch = _torch()
+        _check(self.lib, self.lib.dfann_train(
+            self.h, xt.shape[0], ctypes.c_void_p(xt.data_ptr()),
+            self._stream(torch)))
+        torch.cuda.synchronize()
+
     def add(self, x):
         torch = _torch()
         xt = torch.as_tensor(np.ascontiguousarray(x, dtype=np.float32)).cuda()
@@ -262,6 +269,22 @@ class HipEngine:
             self.h, vmin.ctypes.data_as(ctypes.c_void_p),
             vdiff.ctypes.data_as(ctypes.c_void_p)))
         return vmin, vdiff
+
+    def get_lists(self):
+        """(off, ids, codes) — finalized inverted lists on host; codes are
+        stride-padded rows (stride = dfann_code_stride)."""
+        self.lib.dfann_get_lists.argtypes = [ctypes.c_void_p] + [ctypes.c_void_p] * 3
+        self.lib.dfann_code_stride.argtypes = [ctypes.c_void_p]
+        stride = int(self.lib.dfann_code_stride(self.h))
+        n = self.ntotal
+        off = np.empty(self.nlist + 1, dtype=np.int64)
+        ids = np.empty(max(n, 1), dtype=np.int64)
+        codes = np.empty((max(n, 1), stride), dtype=np.uint8)
+        _check(self.lib, self.lib.dfann_get_lists(
+            self.h, off.ctypes.data_as(ctypes.c_void_p),
+            ids.ctypes.data_as(ctypes.c_void_p),
+            codes.ctypes.data_as(ctypes.c_void_p)))
+        return off, ids[:n], codes[:n]
 
     def set_trained(self, centroids, codebooks=None, vmin=None, vdiff=None):
         def ptr(a):

@@ -107,6 +107,13 @@ int dfann_set_trained(dfann_index *h, const float *centroids_host,
                       const float *codebooks_host, const float *vmin_host,
                       const float *vdiff_host);
 int dfann_get_codebooks(dfann_index *h, float *out_host); /* (m,256,dsub) */
+/* Dump the finalized inverted lists to host: off (nlist+1 i64), ids
+ * (ntotal i64, arrival ids in CSR order), codes (ntotal * stride u8).
+ * stride = code_bytes rounded up to 16. Used by the CPU-baseline leg of
+ * bench.py to scan the SAME index content the GPU holds. */
+int dfann_get_lists(dfann_index *h, int64_t *off_host, int64_t *ids_host,
+                    uint8_t *codes_host);
+int dfann_code_stride(dfann_index *h);
 int dfann_get_sq_params(dfann_index *h, float *vmin_host,
                         float *vdiff_host); /* (d,), (d,) */
 

@@ -307,6 +307,41 @@ class _OracleIVFBase(_OracleIndexBase):
         rows = np.arange(q.shape[0])[:, None]
         return probes, sc[rows, probes]
 
+    # -- common search plumbing (per-class _scan_one(qi, li, bias)) --------
+    # bias: None -> compute internally (seq_ip of q and the list centroid,
+    # IP metric only); else the externally supplied coarse bias (parity
+    # tests feed the engine's coarse output so the scan compares
+    # bit-for-bit regardless of coarse rounding).
+
+    def search(self, q, k):
+        q = self._check_q(q)
+        probes, _ = self.coarse_topn(q, self.nprobe)
+        return self._search_probes(q, probes, None, k)
+
+    def search_preassigned(self, q, probes, keys, k):
+        """probes: (nq, nprobe) list ids; keys: coarse minimize-keys
+        (IP bias = -key), may be None for L2."""
+        q = self._check_q(q)
+        probes = np.asarray(probes)
+        keys = None if keys is None else np.asarray(keys, dtype=np.float32)
+        return self._search_probes(q, probes, keys, k)
+
+    def _search_probes(self, q, probes, keys, k):
+        cand_d, cand_i = [], []
+        for i in range(q.shape[0]):
+            ds, ids = [], []
+            for pi in range(probes.shape[1]):
+                li = int(probes[i, pi])
+                bias = None if keys is None else np.float32(-keys[i, pi])
+                res = self._scan_one(q[i], li, bias)
+                if res is None:
+                    continue
+                ds.append(res[0])
+                ids.append(res[1])
+            cand_d.append(ds)
+            cand_i.append(ids)
+        return self._merge_candidates(cand_d, cand_i, k)
+
     def _merge_candidates(self, cand_d, cand_i, k):
         """cand_*: per-query lists of fp32/int64 arrays -> (D, I)."""
         nq = len(cand_d)
@@ -348,24 +383,13 @@ class OracleIVFFlat(_OracleIVFBase):
                 )
         self.ntotal += x.shape[0]
 
-    def search(self, q, k):
-        q = self._check_q(q)
-        probes, _ = self.coarse_topn(q, self.nprobe)
-        cand_d, cand_i = [], []
-        for i in range(q.shape[0]):
-            ds, ids = [], []
-            for li in probes[i]:
-                xs = self.list_data[li]
-                if xs.shape[0] == 0:
-                    continue
-                if self.metric == METRIC_L2:
-                    ds.append(seq_l2(q[i], xs))
-                else:
-                    ds.append(seq_ip(q[i], xs))
-                ids.append(self.list_ids[li])
-            cand_d.append(ds)
-            cand_i.append(ids)
-        return self._merge_candidates(cand_d, cand_i, k)
+    def _scan_one(self, qi, li, bias):
+        xs = self.list_data[li]
+        if xs.shape[0] == 0:
+            return None
+        if self.metric == METRIC_L2:
+            return seq_l2(qi, xs), self.list_ids[li]
+        return seq_ip(qi, xs), self.list_ids[li]
 
     def reconstruct_ids(self, I):
         """Gather stored vectors for global ids (for search_and_reconstruct)."""
@@ -436,10 +460,17 @@ class OracleIVFPQ(_OracleIVFBase):
         assign = assign_batch(x, self.centroids, self.metric)
         resid = x - self.centroids[assign]
         codes = np.empty((x.shape[0], self.m), dtype=np.uint8)
+        # sequential-accumulation subspace distances (mirrors k_pq_encode:
+        # fp32, mul+add, t ascending — not the BLAS decomposition), chunked
         for j in range(self.m):
             sub = resid[:, j * self.dsub : (j + 1) * self.dsub]
-            sc = pairwise_scores(sub, self.codebooks[j], METRIC_L2)
-            codes[:, j] = np.argmin(sc, axis=1).astype(np.uint8)
+            for s in range(0, x.shape[0], 65536):
+                ss = sub[s : s + 65536]
+                acc = np.zeros((ss.shape[0], 256), dtype=np.float32)
+                for t in range(self.dsub):
+                    diff = ss[:, t, None] - self.codebooks[j][None, :, t]
+                    acc = acc + diff * diff
+                codes[s : s + 65536, j] = np.argmin(acc, axis=1).astype(np.uint8)
         return assign, codes
 
     def add(self, x):
@@ -479,25 +510,15 @@ class OracleIVFPQ(_OracleIVFBase):
         bias = np.float32(seq_ip(qi, self.centroids[li][None, :])[0])
         return lut, bias
 
-    def search(self, q, k):
-        q = self._check_q(q)
-        probes, _ = self.coarse_topn(q, self.nprobe)
-        cand_d, cand_i = [], []
-        for i in range(q.shape[0]):
-            ds, ids = [], []
-            for li in probes[i]:
-                codes = self.list_codes[li]
-                if codes.shape[0] == 0:
-                    continue
-                lut, bias = self.build_lut(q[i], li)
-                dist = adc_scan(lut, codes)
-                if self.metric == METRIC_INNER_PRODUCT:
-                    dist = bias + dist
-                ds.append(dist)
-                ids.append(self.list_ids[li])
-            cand_d.append(ds)
-            cand_i.append(ids)
-        return self._merge_candidates(cand_d, cand_i, k)
+    def _scan_one(self, qi, li, bias):
+        codes = self.list_codes[li]
+        if codes.shape[0] == 0:
+            return None
+        lut, bias0 = self.build_lut(qi, li)
+        dist = adc_scan(lut, codes)
+        if self.metric == METRIC_INNER_PRODUCT:
+            dist = (bias0 if bias is None else bias) + dist
+        return dist, self.list_ids[li]
 
     def decode_ids(self, I):
         lut = {}
@@ -598,27 +619,17 @@ class OracleIVFSQ(_OracleIVFBase):
                 )
         self.ntotal += x.shape[0]
 
-    def search(self, q, k):
-        q = self._check_q(q)
-        probes, _ = self.coarse_topn(q, self.nprobe)
-        cand_d, cand_i = [], []
-        for i in range(q.shape[0]):
-            ds, ids = [], []
-            for li in probes[i]:
-                codes = self.list_codes[li]
-                if codes.shape[0] == 0:
-                    continue
-                dec = self._decode_codes(codes)  # residual values
-                if self.metric == METRIC_L2:
-                    r = (q[i] - self.centroids[li]).astype(np.float32)
-                    ds.append(seq_l2(r, dec))
-                else:
-                    bias = np.float32(seq_ip(q[i], self.centroids[li][None, :])[0])
-                    ds.append(bias + seq_ip(q[i], dec))
-                ids.append(self.list_ids[li])
-            cand_d.append(ds)
-            cand_i.append(ids)
-        return self._merge_candidates(cand_d, cand_i, k)
+    def _scan_one(self, qi, li, bias):
+        codes = self.list_codes[li]
+        if codes.shape[0] == 0:
+            return None
+        dec = self._decode_codes(codes)  # residual values
+        if self.metric == METRIC_L2:
+            r = (qi - self.centroids[li]).astype(np.float32)
+            return seq_l2(r, dec), self.list_ids[li]
+        if bias is None:
+            bias = np.float32(seq_ip(qi, self.centroids[li][None, :])[0])
+        return bias + seq_ip(qi, dec), self.list_ids[li]
 
     def decode_ids(self, I):
         lut = {}
