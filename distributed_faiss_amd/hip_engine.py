@@ -40,6 +40,14 @@ def load_lib():
     global _LIB
     if _LIB is not None:
         return _LIB
+    # Load torch FIRST so its bundled libamdhip64 owns the soname: loading
+    # libdfann.so first binds the system ROCm runtime, which on the GPU
+    # boxes reports "no ROCm-capable device" while torch's runtime works
+    # (two HIP runtimes in one process otherwise).
+    try:
+        import torch  # noqa: F401
+    except ImportError:
+        pass
     p = lib_path()
     if not os.path.exists(p):
         raise RuntimeError(
