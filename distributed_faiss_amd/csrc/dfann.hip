@@ -42,6 +42,23 @@ static thread_local std::string g_err;
 
 extern "C" const char *dfann_last_error(void) { return g_err.c_str(); }
 
+// DFANN_TRACE=1: synchronize + log after each build/search phase (debug)
+static bool trace_on() {
+  static int v = -1;
+  if (v < 0) {
+    const char *e = getenv("DFANN_TRACE");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+
+static void trace_point(const char *what, hipStream_t s) {
+  if (!trace_on()) return;
+  hipError_t e = hipStreamSynchronize(s);
+  fprintf(stderr, "[dfann] %s: %s\n", what, hipGetErrorString(e));
+  fflush(stderr);
+}
+
 // ---------------------------------------------------------------------------
 // tiny JSON (flat dict of scalars; produced by our own Python side)
 // ---------------------------------------------------------------------------
@@ -88,6 +105,10 @@ static std::string json_str(const std::string &js, const char *key,
 struct DevBuf {
   void *p = nullptr;
   size_t cap = 0;
+  DevBuf() = default;
+  DevBuf(const DevBuf &) = delete;
+  DevBuf &operator=(const DevBuf &) = delete;
+  ~DevBuf() { free(); }
   void ensure(size_t bytes) {
     if (bytes <= cap) return;
     void *np = nullptr;
@@ -266,7 +287,7 @@ static void assign_rows(dfann_index *h, const float *x, int64_t n,
               h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, keys,
               stream);
     hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream, keys, c,
-                       (long long)nlist, (long long)nlist, 0, bestv,
+                       (long long)nlist, (long long)nlist, 0, bestv + s,
                        assign_dev + s);
   }
   HIP_CHECK(hipGetLastError());
@@ -294,6 +315,7 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
     xt = xt_buf.as<float>();
     nt = cap;
   }
+  trace_point("kmeans:subsample", stream);
   if (nt < kcent) throw std::runtime_error("kmeans: n < k");
 
   // init centroids
@@ -305,6 +327,7 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
   hipLaunchKernelGGL(k_gather_rows, grid1d((int64_t)kcent * d), dim3(256), 0,
                      stream, xt, idx_buf.as<int64_t>(), (long long)kcent, d,
                      cent_out);
+  trace_point("kmeans:init", stream);
 
   DevBuf cn, asg, bestv, keys, sums, counts;
   cn.ensure((size_t)kcent * 4);
@@ -328,14 +351,16 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
                 nullptr, metric == M_IP ? 0 : 1, keys.as<float>(), stream);
       hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream,
                          keys.as<float>(), c, (long long)kcent,
-                         (long long)kcent, 0, bestv.as<float>(),
+                         (long long)kcent, 0, bestv.as<float>() + s,
                          asg.as<int>() + s);
     }
     HIP_CHECK(hipMemsetAsync(sums.p, 0, (size_t)kcent * d * 4, stream));
     HIP_CHECK(hipMemsetAsync(counts.p, 0, (size_t)kcent * 4, stream));
+    trace_point("kmeans:assign", stream);
     hipLaunchKernelGGL(k_centroid_accum, grid1d(nt * d), dim3(256), 0, stream,
                        xt, asg.as<int>(), nt, d, sums.as<float>(),
                        counts.as<int>());
+    trace_point("kmeans:accum", stream);
     hipLaunchKernelGGL(k_centroid_div, grid1d((int64_t)kcent * d), dim3(256), 0,
                        stream, cent_out, sums.as<float>(), counts.as<int>(),
                        (long long)kcent, d);
@@ -422,8 +447,10 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
   if (h->trained) return;  // faiss semantics: re-train of trained is a no-op here
   h->centroids.ensure((size_t)h->nlist * h->d * 4);
   h->cnorm.ensure((size_t)h->nlist * 4);
+  trace_point("train:begin", stream);
   kmeans_device(h, x, n, h->nlist, h->d, h->metric, h->seed,
                 h->centroids.as<float>(), stream);
+  trace_point("train:coarse-kmeans", stream);
   rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(),
            stream);
   if (h->type == T_IVFPQ || (h->type == T_IVFSQ && h->sq8)) {
@@ -431,9 +458,11 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
     asg.ensure((size_t)n * 4);
     resid.ensure((size_t)n * h->d * 4);
     assign_rows(h, x, n, asg.as<int>(), stream);
+    trace_point("train:assign", stream);
     hipLaunchKernelGGL(k_residual, grid1d(n * h->d), dim3(256), 0, stream, x,
                        h->centroids.as<float>(), asg.as<int>(), n, h->d,
                        resid.as<float>());
+    trace_point("train:residual", stream);
     if (h->type == T_IVFPQ) {
       h->codebooks.ensure((size_t)h->m * 256 * h->dsub * 4);
       DevBuf sub;
@@ -446,6 +475,7 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
                       h->seed + 1 + j,
                       h->codebooks.as<float>() + (size_t)j * 256 * h->dsub,
                       stream);
+        trace_point("train:pq-sub", stream);
       }
     } else {  // SQ8 ranges on residuals
       DevBuf mn, mx;
