@@ -189,6 +189,8 @@ struct dfann_index {
   bool trained = false;
   int64_t ntotal = 0;
   int code_bytes = 0, stride = 0;
+  int ws_mb = 512;  // chunk budget for key matrices (spec "ws_mb"; tests
+                    // shrink it to force the multi-chunk paths)
 
   DevBuf centroids, cnorm, codebooks, sq_vmin, sq_vdiff, sq_scale;
   // staging (arrival order)
@@ -272,8 +274,8 @@ static void rownorms(const float *x, int64_t n, int d, float *out,
 static void assign_rows(dfann_index *h, const float *x, int64_t n,
                         int32_t *assign_dev, hipStream_t stream) {
   int nlist = h->nlist;
-  // chunk points so the key matrix stays <= ~512 MB
-  int64_t chunk = std::max<int64_t>(1, (512LL << 20) / ((int64_t)nlist * 4));
+  // chunk points so the key matrix stays within the ws budget
+  int64_t chunk = std::max<int64_t>(1, ((int64_t)h->ws_mb << 20) / ((int64_t)nlist * 4));
   chunk = std::min<int64_t>(chunk, n);
   h->ws1.ensure((size_t)chunk * nlist * 4);
   float *keys = h->ws1.as<float>();
@@ -335,7 +337,8 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
   bestv.ensure((size_t)nt * 4);
   sums.ensure((size_t)kcent * d * 4);
   counts.ensure((size_t)kcent * 4);
-  int64_t chunk = std::max<int64_t>(1, (512LL << 20) / ((int64_t)kcent * 4));
+  int64_t chunk = std::max<int64_t>(
+      1, ((int64_t)(h ? h->ws_mb : 512) << 20) / ((int64_t)kcent * 4));
   chunk = std::min(chunk, nt);
   keys.ensure((size_t)chunk * kcent * 4);
   std::vector<int> h_counts(kcent);
@@ -416,6 +419,8 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->nprobe = (int)json_int(js, "nprobe", 1);
   h->seed = (uint64_t)json_int(js, "seed", 1234);
   h->sq8 = json_str(js, "sq_type", "fp16") == "8bit";
+  h->ws_mb = (int)json_int(js, "ws_mb", 512);
+  if (h->ws_mb < 1) h->ws_mb = 1;
   if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
   if (h->type != T_FLAT && h->nlist <= 0) {
     delete h;
@@ -601,7 +606,7 @@ static void pad_fill(dfann_index *h, int64_t nq, int k, float *D, int64_t *I,
 static void coarse_impl(dfann_index *h, int64_t nq, const float *q, int nprobe,
                         int32_t *probes, float *keys, hipStream_t stream) {
   int nlist = h->nlist;
-  int64_t chunk = std::max<int64_t>(1, (512LL << 20) / ((int64_t)nlist * 4));
+  int64_t chunk = std::max<int64_t>(1, ((int64_t)h->ws_mb << 20) / ((int64_t)nlist * 4));
   chunk = std::min<int64_t>(chunk, nq);
   h->ws1.ensure((size_t)chunk * nlist * 4);
   float *sc = h->ws1.as<float>();
@@ -689,10 +694,11 @@ static void flat_search_impl(dfann_index *h, int64_t nq, const float *q, int k,
                              float *D, int64_t *I, hipStream_t stream) {
   if (h->ntotal == 0) { pad_fill(h, nq, k, D, I, stream); return; }
   bool ip = h->metric == M_IP;
-  const int64_t CH = 65536;
+  const int64_t CH = std::min<int64_t>(65536, std::max<int64_t>(
+      4096, ((int64_t)h->ws_mb << 20) / (64 * 4)));
   int64_t nch = (h->ntotal + CH - 1) / CH;
   // chunk queries so scores fit
-  int64_t qch = std::max<int64_t>(1, (512LL << 20) / (CH * 4));
+  int64_t qch = std::max<int64_t>(1, ((int64_t)h->ws_mb << 20) / (CH * 4));
   qch = std::min(qch, nq);
   h->ws1.ensure((size_t)qch * CH * 4);
   h->ws2.ensure((size_t)std::max<int64_t>(CH, nq) * 4);            // bnorm / qnorm

@@ -233,6 +233,39 @@ def test_ivfflat_tolerance_parity():
     np.testing.assert_allclose(D[same], Do[same], rtol=1e-4, atol=1e-4)
 
 
+def test_chunked_assign_matches_unchunked():
+    # ws_mb=1 forces the multi-chunk assign/coarse paths (the round-1
+    # negative-OOB regression lived there): results must be identical to
+    # the single-chunk engine
+    d, n = 32, 30000
+    xb, q = _rand(n, d, 40), _rand(20, d, 41)
+    specs = [dict(type="ivf_flat", dim=d, metric=L2, nlist=64, nprobe=64,
+                  seed=7, ws_mb=w) for w in (1, 512)]
+    res = []
+    for spec in specs:
+        eng = HipEngine(spec=spec)
+        eng.train(xb)
+        eng.add(xb)
+        res.append(eng.search(q, 10))
+    np.testing.assert_array_equal(res[0][1], res[1][1])
+    np.testing.assert_array_equal(res[0][0], res[1][0])
+
+
+def test_flat_multichunk_matches_oracle():
+    # > one 65536-column chunk: chunk-winner merge path
+    d, n = 16, 150000
+    xb, q = _rand(n, d, 42), _rand(9, d, 43)
+    eng = HipEngine(spec={"type": "flat", "dim": d, "metric": L2})
+    eng.train(xb)
+    eng.add(xb)
+    D, I = eng.search(q, 7)
+    orc = make_oracle_engine({"type": "flat", "dim": d, "metric": L2})
+    orc.add(xb)
+    Do, Io = orc.search(q, 7)
+    assert (I == Io).mean() > 0.999
+    np.testing.assert_allclose(D, Do, rtol=1e-4, atol=1e-4)
+
+
 # ---------------------------------------------------------------------------
 # decode / persistence / merge
 # ---------------------------------------------------------------------------
