@@ -39,10 +39,12 @@ def log(msg):
 
 
 WORKLOADS = {
-    # BASELINE.json configs[2] (SIFT1M-shaped IVFPQ)
+    # BASELINE.json configs[2] (SIFT1M-shaped IVFPQ). sigma 0.5 gives
+    # SIFT-like cluster overlap so the nprobe/recall trade-off is
+    # non-degenerate (sigma 0.15 was solved at nprobe=1).
     "ivfpq_1m_d128_m16": dict(
         type="ivfpq", d=128, n=1_000_000, nlist=1024, m=16, nbits=8,
-        metric=1, nq=10_000, k=10, centers=10_000, sigma=0.15,
+        metric=1, nq=10_000, k=10, centers=10_000, sigma=0.5,
     ),
     # BASELINE.json configs[1] (ivf_simple 1M, dot) — parity/regression
     "ivfflat_1m_d128": dict(
@@ -277,18 +279,28 @@ def main():
         orc.ntotal = eng.ntotal
         orc.nprobe = nprobe
         log(f"cpu baseline index shared in {time.time()-t0:.1f}s")
+        # bounded sample: repeat query slices until >= ~10 s of CPU work
         nq_s = args.cpu_sample_queries
         q_np = q[:nq_s].cpu().numpy()
+        done = 0
         t0 = time.time()
-        orc.search(q_np, k)
-        cpu_elapsed = time.time() - t0
+        while True:
+            orc.search(q_np, k)
+            done += nq_s
+            cpu_elapsed = time.time() - t0
+            if cpu_elapsed >= 10.0 or done >= cfg["nq"]:
+                break
         cpu_baseline = {
-            "value": nq_s / cpu_elapsed,
+            "value": done / cpu_elapsed,
             "unit": "QPS",
-            "cores": os.cpu_count(),
+            # numpy fancy-indexing scan is single-threaded; the BLAS coarse
+            # GEMM uses all cores but is a small fraction of oracle time
+            "cores": 1,
             "kind": "port",
-            "sample": f"{nq_s} queries of the same batch, same index "
-                      f"content (engine lists), {cpu_elapsed:.1f}s",
+            "sample": f"{done} queries ({nq_s}-query batches of the same "
+                      f"workload batch), same index content (engine lists), "
+                      f"{cpu_elapsed:.1f}s on host cores "
+                      f"(os.cpu_count={os.cpu_count()})",
         }
 
     if rank == 0:

@@ -235,6 +235,8 @@ struct dfann_index {
 // launch helpers
 // ---------------------------------------------------------------------------
 
+static bool use_regsel(int k) { return k <= 16; }
+
 static dim3 grid1d(int64_t total, int block = 256, int64_t cap = 65535LL * 8) {
   int64_t g = (total + block - 1) / block;
   if (g > cap) g = cap;
@@ -615,10 +617,19 @@ static void coarse_impl(dfann_index *h, int64_t nq, const float *q, int nprobe,
     gemm_keys(h, q + s * h->d, c, h->centroids.as<float>(), nlist, h->d,
               h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, sc,
               stream);
-    hipLaunchKernelGGL(k_topk_rows, dim3((unsigned)c), dim3(256), SEL_LDS_BYTES,
-                       stream, sc, c, (long long)nlist, (long long)nlist,
-                       nprobe, 0u, (long long)nprobe, keys + s * nprobe,
-                       reinterpret_cast<unsigned *>(probes) + s * nprobe);
+    if (use_regsel(nprobe)) {
+      hipLaunchKernelGGL(k_topk_rows_rk, dim3((unsigned)c), dim3(256),
+                         REGSEL_LDS_BYTES, stream, sc, c, (long long)nlist,
+                         (long long)nlist, nprobe, 0u, (long long)nprobe,
+                         keys + s * nprobe,
+                         reinterpret_cast<unsigned *>(probes) + s * nprobe);
+    } else {
+      hipLaunchKernelGGL(k_topk_rows, dim3((unsigned)c), dim3(256),
+                         SEL_LDS_BYTES, stream, sc, c, (long long)nlist,
+                         (long long)nlist, nprobe, 0u, (long long)nprobe,
+                         keys + s * nprobe,
+                         reinterpret_cast<unsigned *>(probes) + s * nprobe);
+    }
   }
   HIP_CHECK(hipGetLastError());
 }
@@ -638,28 +649,33 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                const int64_t *, int, int, int, int, int, int, int, float *,
                unsigned *, int) = nullptr;
   bool ip = h->metric == M_IP;
+  bool rk = use_regsel(k);
   switch (h->type) {
     case T_IVFPQ:
       fam_floats = h->m * 256 + h->d;
-      kern = ip ? k_scan_pq_ip : k_scan_pq_l2;
+      kern = rk ? (ip ? k_scan_pq_ip_rk : k_scan_pq_l2_rk)
+                : (ip ? k_scan_pq_ip : k_scan_pq_l2);
       break;
     case T_IVFFLAT:
       fam_floats = h->d;
-      kern = ip ? k_scan_ivfflat_ip : k_scan_ivfflat_l2;
+      kern = rk ? (ip ? k_scan_ivfflat_ip_rk : k_scan_ivfflat_l2_rk)
+                : (ip ? k_scan_ivfflat_ip : k_scan_ivfflat_l2);
       break;
     case T_IVFSQ:
       if (h->sq8) {
         fam_floats = 3 * h->d;
-        kern = ip ? k_scan_sq8_ip : k_scan_sq8_l2;
+        kern = rk ? (ip ? k_scan_sq8_ip_rk : k_scan_sq8_l2_rk)
+                  : (ip ? k_scan_sq8_ip : k_scan_sq8_l2);
       } else {
         fam_floats = h->d;
-        kern = ip ? k_scan_sqf_ip : k_scan_sqf_l2;
+        kern = rk ? (ip ? k_scan_sqf_ip_rk : k_scan_sqf_l2_rk)
+                  : (ip ? k_scan_sqf_ip : k_scan_sqf_l2);
       }
       break;
     default:
       throw std::runtime_error("scan on flat index");
   }
-  size_t lds = (size_t)fam_floats * 4 + SEL_LDS_BYTES;
+  size_t lds = (size_t)fam_floats * 4 + (rk ? REGSEL_LDS_BYTES : SEL_LDS_BYTES);
   if (lds > 160 * 1024)
     throw std::runtime_error("scan LDS over budget (m too large)");
   TimingEv e;
@@ -683,9 +699,15 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   }
   TimingEv em;
   if (h->timing) em = h->ev_begin(stream);
-  hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256), SEL_LDS_BYTES,
-                     stream, cand_d, cand_p, nq, nprobe * k, k,
-                     h->cr_ids.as<int64_t>(), ip ? 1 : 0, D, I);
+  if (use_regsel(k)) {
+    hipLaunchKernelGGL(k_merge_cand_rk, dim3((unsigned)nq), dim3(256),
+                       REGSEL_LDS_BYTES + 128, stream, cand_d, cand_p, nq,
+                       nprobe * k, k, h->cr_ids.as<int64_t>(), ip ? 1 : 0, D, I);
+  } else {
+    hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256),
+                       SEL_LDS_BYTES, stream, cand_d, cand_p, nq, nprobe * k, k,
+                       h->cr_ids.as<int64_t>(), ip ? 1 : 0, D, I);
+  }
   if (h->timing) h->ev_end(em, stream, h->ev_merge);
   HIP_CHECK(hipGetLastError());
 }
@@ -721,17 +743,32 @@ static void flat_search_impl(dfann_index *h, int64_t nq, const float *q, int k,
       gemm_keys(h, q + s * h->d, c, base, bn_rows, h->d, bn, qn + s,
                 ip ? 0 : 2, sc, stream);
       // winners layout: (nq, nch, k) — row stride nch*k
-      hipLaunchKernelGGL(k_topk_rows, dim3((unsigned)c), dim3(256),
-                         SEL_LDS_BYTES, stream, sc, c, (long long)bn_rows,
-                         (long long)bn_rows, k, (unsigned)b0,
-                         (long long)(nch * k), wd + (s * nch + ci) * k,
-                         wp + (s * nch + ci) * k);
+      if (use_regsel(k)) {
+        hipLaunchKernelGGL(k_topk_rows_rk, dim3((unsigned)c), dim3(256),
+                           REGSEL_LDS_BYTES, stream, sc, c, (long long)bn_rows,
+                           (long long)bn_rows, k, (unsigned)b0,
+                           (long long)(nch * k), wd + (s * nch + ci) * k,
+                           wp + (s * nch + ci) * k);
+      } else {
+        hipLaunchKernelGGL(k_topk_rows, dim3((unsigned)c), dim3(256),
+                           SEL_LDS_BYTES, stream, sc, c, (long long)bn_rows,
+                           (long long)bn_rows, k, (unsigned)b0,
+                           (long long)(nch * k), wd + (s * nch + ci) * k,
+                           wp + (s * nch + ci) * k);
+      }
     }
   }
   // merge chunk winners
-  hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256), SEL_LDS_BYTES,
-                     stream, wd, wp, nq, (int)(nch * k), k,
-                     (const int64_t *)nullptr, ip ? 1 : 0, D, I);
+  if (use_regsel(k)) {
+    hipLaunchKernelGGL(k_merge_cand_rk, dim3((unsigned)nq), dim3(256),
+                       REGSEL_LDS_BYTES + 128, stream, wd, wp, nq,
+                       (int)(nch * k), k, (const int64_t *)nullptr,
+                       ip ? 1 : 0, D, I);
+  } else {
+    hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256),
+                       SEL_LDS_BYTES, stream, wd, wp, nq, (int)(nch * k), k,
+                       (const int64_t *)nullptr, ip ? 1 : 0, D, I);
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -963,9 +1000,15 @@ extern "C" int dfann_merge_topk(int64_t nq, int S, int k, const float *D_dev,
   API_BEGIN
   (void)I_dev;  // slots map to metadata host-side (ref client.py:297-298)
   if (k > 512) throw std::runtime_error("k > 512 unsupported");
-  hipLaunchKernelGGL(k_merge_shards, dim3((unsigned)nq), dim3(256),
-                     SEL_LDS_BYTES, (hipStream_t)stream, D_dev, nq, S, k,
-                     maximize, Dout_dev, Iout_dev);
+  if (use_regsel(k)) {
+    hipLaunchKernelGGL(k_merge_shards_rk, dim3((unsigned)nq), dim3(256),
+                       REGSEL_LDS_BYTES + 128, (hipStream_t)stream, D_dev, nq,
+                       S, k, maximize, Dout_dev, Iout_dev);
+  } else {
+    hipLaunchKernelGGL(k_merge_shards, dim3((unsigned)nq), dim3(256),
+                       SEL_LDS_BYTES, (hipStream_t)stream, D_dev, nq, S, k,
+                       maximize, Dout_dev, Iout_dev);
+  }
   HIP_CHECK(hipGetLastError());
   API_END
 }

@@ -125,6 +125,95 @@ __device__ __forceinline__ void sel_guard(Sel s, int k, int max_appends) {
 }
 
 // ---------------------------------------------------------------------------
+// register-resident per-thread top-K (K <= 16) — the fast selection path.
+// Each thread keeps an ascending (dist, pos) array in registers (static
+// indices via full unroll); most elements fail the single arr[K-1] compare.
+// Block-wide result: K extraction rounds over the threads' heads.
+// ---------------------------------------------------------------------------
+
+template <int K>
+struct RegTopK {
+  float d[K];
+  unsigned p[K];
+
+  __device__ __forceinline__ void init() {
+#pragma unroll
+    for (int i = 0; i < K; ++i) {
+      d[i] = DFANN_FLT_MAX;
+      p[i] = PAD_POS;
+    }
+  }
+
+  __device__ __forceinline__ void push(float nd, unsigned np) {
+    if (!sel_less(nd, np, d[K - 1], p[K - 1])) return;
+    d[K - 1] = nd;
+    p[K - 1] = np;
+#pragma unroll
+    for (int i = K - 1; i > 0; --i) {
+      if (sel_less(d[i], p[i], d[i - 1], p[i - 1])) {
+        float td = d[i]; d[i] = d[i - 1]; d[i - 1] = td;
+        unsigned tp = p[i]; p[i] = p[i - 1]; p[i - 1] = tp;
+      }
+    }
+  }
+
+  // drop the current head (after it won an extraction round)
+  __device__ __forceinline__ void pop_head() {
+#pragma unroll
+    for (int i = 0; i < K - 1; ++i) {
+      d[i] = d[i + 1];
+      p[i] = p[i + 1];
+    }
+    d[K - 1] = DFANN_FLT_MAX;
+    p[K - 1] = PAD_POS;
+  }
+};
+
+// Block-wide merge of per-thread RegTopK heads: k extraction rounds.
+// lds: 48 bytes of scratch. out_d/out_p: k contiguous results (ascending).
+#define REGSEL_LDS_BYTES 64
+template <int K>
+__device__ void regtopk_block_extract(RegTopK<K> &loc, int k, char *lds,
+                                      float *out_d, unsigned *out_p) {
+  float *wd = reinterpret_cast<float *>(lds);           // per-wave best dist
+  unsigned *wp = reinterpret_cast<unsigned *>(lds + 16); // per-wave best pos
+  unsigned *wl = reinterpret_cast<unsigned *>(lds + 32); // per-wave best lane
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int nw = blockDim.x >> 6;
+  for (int round = 0; round < k; ++round) {
+    float cd = loc.d[0];
+    unsigned cp = loc.p[0];
+    unsigned cl = threadIdx.x;
+    // wave reduction (keep the smaller (d,p))
+#pragma unroll
+    for (int o = 32; o > 0; o >>= 1) {
+      float od = __shfl_down(cd, o, 64);
+      unsigned op = __shfl_down(cp, o, 64);
+      unsigned ol = __shfl_down(cl, o, 64);
+      if (sel_less(od, op, cd, cp)) { cd = od; cp = op; cl = ol; }
+    }
+    if (lane == 0) { wd[w] = cd; wp[w] = cp; wl[w] = cl; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float bd = wd[0]; unsigned bp = wp[0], bl = wl[0];
+      for (int i = 1; i < nw; ++i)
+        if (sel_less(wd[i], wp[i], bd, bp)) { bd = wd[i]; bp = wp[i]; bl = wl[i]; }
+      wd[0] = bd; wp[0] = bp; wl[0] = bl;
+    }
+    __syncthreads();
+    float bd = wd[0];
+    unsigned bp = wp[0], bl = wl[0];
+    if (threadIdx.x == 0) {
+      bool valid = bp != PAD_POS;
+      out_d[round] = valid ? bd : DFANN_FLT_MAX;
+      out_p[round] = valid ? bp : PAD_POS;
+    }
+    if (threadIdx.x == bl) loc.pop_head();
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // GEMM (fp32 MFMA): C[i][j] = sum_k A[i][k] * B[j][k]   (A: MxK, B: NxK)
 // 128x128 tile, 4 waves x (2x2 of 32x32) on v_mfma_f32_32x32x2_f32.
 // Operand map (cdna_hip_programming.md §3): lane l feeds A[i=l&31][k=l>>5],
@@ -266,6 +355,24 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows(
   }
 }
 
+// register-path top-k per row (k <= 16): no LDS buffer, no bitonic
+extern "C" __global__ __launch_bounds__(256) void k_topk_rows_rk(
+    const float *__restrict__ keys, long long rows, long long cols,
+    long long ldk, int k, unsigned base, long long ldo,
+    float *__restrict__ out_d, unsigned *__restrict__ out_p) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  long long row = blockIdx.x;
+  if (row >= rows) return;
+  RegTopK<16> loc;
+  loc.init();
+  const float *kp = keys + row * ldk;
+  for (long long c = threadIdx.x; c < cols; c += blockDim.x)
+    loc.push(kp[c], (unsigned)c + base);
+  __syncthreads();
+  regtopk_block_extract<16>(loc, k, smem, out_d + row * ldo,
+                            out_p + row * ldo);
+}
+
 // ---------------------------------------------------------------------------
 // running argmin across key-matrix chunks (assignment). rows = points.
 // best_v/best_i persist across chunk calls (init by k_fill_assign_init).
@@ -306,7 +413,74 @@ extern "C" __global__ void k_assign_chunk(const float *__restrict__ keys,
 // keys_dev: coarse minimize-keys (IP bias = -key); null for L2.
 // ---------------------------------------------------------------------------
 
+// per-row distance (shared by both selection paths); fam = staged LDS
+// region (FAM 0: LUT; FAM 2: [target][vmin][scale]; FAM 3: target).
 template <int FAM, bool IS_IP>
+__device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
+                                               const float *__restrict__ fam,
+                                               int d, int m) {
+  float acc = 0.f;
+  if (FAM == 0) {
+    const float *lut = fam;
+    for (int g = 0; g < m; g += 16) {
+#pragma clang fp contract(off)
+      uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
+      unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
+#pragma unroll
+      for (int b = 0; b < 16; ++b) {
+        if (g + b < m) {
+          unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
+          unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
+          acc = acc + lut[(g + b) * 256 + c];
+        }
+      }
+    }
+  } else if (FAM == 2) {
+    const float *rbuf = fam, *vm = fam + d, *sc = fam + 2 * d;
+    for (int g = 0; g < d; g += 16) {
+#pragma clang fp contract(off)
+      uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
+      unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
+#pragma unroll
+      for (int b = 0; b < 16; ++b) {
+        if (g + b < d) {
+          unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
+          unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
+          int t = g + b;
+          float dec = vm[t] + ((float)c + 0.5f) * sc[t];
+          if (IS_IP) acc = acc + rbuf[t] * dec;
+          else {
+            float diff = rbuf[t] - dec;
+            acc = acc + diff * diff;
+          }
+        }
+      }
+    }
+  } else {  // FAM 3: fp16 codes
+    const float *rbuf = fam;
+    for (int g = 0; g < d; g += 8) {
+#pragma clang fp contract(off)
+      uint4 wv = *reinterpret_cast<const uint4 *>(cp + (size_t)g * 2);
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        if (g + b < d) {
+          unsigned word = (b < 2) ? wv.x : (b < 4) ? wv.y : (b < 6) ? wv.z : wv.w;
+          unsigned h = (word >> ((b & 1) * 16)) & 0xFFFFu;
+          float dec = __half2float(__ushort_as_half((unsigned short)h));
+          int t = g + b;
+          if (IS_IP) acc = acc + rbuf[t] * dec;
+          else {
+            float diff = rbuf[t] - dec;
+            acc = acc + diff * diff;
+          }
+        }
+      }
+    }
+  }
+  return acc;
+}
+
+template <int FAM, bool IS_IP, bool REGSEL>
 __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
@@ -317,7 +491,7 @@ __device__ void ivf_scan_body(
     unsigned *__restrict__ cand_p, int fam_floats) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float *fam = reinterpret_cast<float *>(smem);
-  Sel s = sel_carve(smem + (size_t)fam_floats * 4);
+  char *selbase = smem + (size_t)fam_floats * 4;
   long long blk = blockIdx.x;
   int bq = (int)(blk / nprobe), bp = (int)(blk % nprobe);
   int L = probes[(long long)bq * nprobe + bp];
@@ -336,7 +510,7 @@ __device__ void ivf_scan_body(
 
   // --- stage FAM region ---
   if (FAM == 0) {
-    // rbuf (d floats) then LUT (m*256) — rbuf lives after the LUT
+    // rbuf (d floats) AFTER the LUT
     float *lut = fam;
     float *rbuf = fam + (size_t)m * 256;
     for (int t = threadIdx.x; t < d; t += blockDim.x)
@@ -371,7 +545,15 @@ __device__ void ivf_scan_body(
     for (int t = threadIdx.x; t < d; t += blockDim.x)
       fam[t] = IS_IP ? qp[t] : qp[t] - cent[(long long)L * d + t];
   }
-  sel_init(s);
+
+  RegTopK<16> loc;
+  Sel s;
+  if (REGSEL) {
+    loc.init();
+  } else {
+    s = sel_carve(selbase);
+    sel_init(s);
+  }
   __syncthreads();
 
   // --- scan ---
@@ -379,7 +561,7 @@ __device__ void ivf_scan_body(
     // 16-lane-per-vector tree reduction (tolerance parity path)
     int sub = threadIdx.x & 15, grp = threadIdx.x >> 4;  // 16 groups
     for (long long base = s0; base < s1; base += 16 * 8) {
-      sel_guard(s, k, 128);
+      if (!REGSEL) sel_guard(s, k, 128);
       for (int u = 0; u < 8; ++u) {
         long long pos = base + (long long)u * 16 + grp;
         float dist = 0.f;
@@ -399,114 +581,72 @@ __device__ void ivf_scan_body(
           for (int o = 8; o > 0; o >>= 1) part += __shfl_xor(part, o, 16);
           dist = IS_IP ? -part : part;
         }
-        if (valid && sub == 0) sel_try(s, dist, (unsigned)pos);
+        if (valid && sub == 0) {
+          if (REGSEL) loc.push(dist, (unsigned)pos);
+          else sel_try(s, dist, (unsigned)pos);
+        }
       }
     }
   } else {
-    const float *lut = fam;                       // FAM 0
-    const float *rbuf = fam;                      // FAM 2/3 target vector
-    const float *vm = fam + d, *sc = fam + 2 * d; // FAM 2
     for (long long base = s0; base < s1; base += 512) {
-      sel_guard(s, k, 512);
+      if (!REGSEL) sel_guard(s, k, 512);
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
         long long pos = base + (long long)u * 256 + threadIdx.x;
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
-          float acc = 0.f;
-          if (FAM == 0) {
-#pragma clang fp contract(off)
-            for (int g = 0; g < m; g += 16) {
-              uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
-              unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
-#pragma unroll
-              for (int b = 0; b < 16; ++b) {
-                if (g + b < m) {
-                  unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
-                  unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
-                  acc = acc + lut[(g + b) * 256 + c];
-                }
-              }
-            }
-          } else if (FAM == 2) {
-#pragma clang fp contract(off)
-            for (int g = 0; g < d; g += 16) {
-              uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
-              unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
-#pragma unroll
-              for (int b = 0; b < 16; ++b) {
-                if (g + b < d) {
-                  unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
-                  unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
-                  int t = g + b;
-                  float dec = vm[t] + ((float)c + 0.5f) * sc[t];
-                  if (IS_IP) acc = acc + rbuf[t] * dec;
-                  else {
-                    float diff = rbuf[t] - dec;
-                    acc = acc + diff * diff;
-                  }
-                }
-              }
-            }
-          } else {  // FAM 3: fp16 codes
-#pragma clang fp contract(off)
-            for (int g = 0; g < d; g += 8) {  // 8 halves = 16 bytes
-              uint4 wv = *reinterpret_cast<const uint4 *>(cp + (size_t)g * 2);
-              unsigned ww[1];
-#pragma unroll
-              for (int b = 0; b < 8; ++b) {
-                if (g + b < d) {
-                  unsigned word = (b < 2) ? wv.x : (b < 4) ? wv.y : (b < 6) ? wv.z : wv.w;
-                  unsigned h = (word >> ((b & 1) * 16)) & 0xFFFFu;
-                  __half hv = __ushort_as_half((unsigned short)h);
-                  float dec = __half2float(hv);
-                  int t = g + b;
-                  if (IS_IP) acc = acc + rbuf[t] * dec;
-                  else {
-                    float diff = rbuf[t] - dec;
-                    acc = acc + diff * diff;
-                  }
-                }
-              }
-              (void)ww;
-            }
-          }
+          float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
           float dist = IS_IP ? -(bias + acc) : acc;
-          sel_try(s, dist, (unsigned)pos);
+          if (REGSEL) loc.push(dist, (unsigned)pos);
+          else sel_try(s, dist, (unsigned)pos);
         }
       }
     }
   }
   __syncthreads();
-  sel_compact(s, k);
-  int cnt = *s.cnt;
-  for (int j = threadIdx.x; j < k; j += blockDim.x) {
-    bool v = j < cnt;
-    cand_d[out_base + j] = v ? s.d[j] : DFANN_FLT_MAX;
-    cand_p[out_base + j] = v ? s.p[j] : PAD_POS;
+  if (REGSEL) {
+    regtopk_block_extract<16>(loc, k, selbase, cand_d + out_base,
+                              cand_p + out_base);
+  } else {
+    sel_compact(s, k);
+    int cnt = *s.cnt;
+    for (int j = threadIdx.x; j < k; j += blockDim.x) {
+      bool v = j < cnt;
+      cand_d[out_base + j] = v ? s.d[j] : DFANN_FLT_MAX;
+      cand_p[out_base + j] = v ? s.p[j] : PAD_POS;
+    }
   }
 }
 
-#define INSTANTIATE_SCAN(NAME, FAM, IS_IP)                                     \
+#define INSTANTIATE_SCAN(NAME, FAM, IS_IP, REGSEL)                             \
   extern "C" __global__ __launch_bounds__(256) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
-      const float *keys, const uint8_t *codes, const int64_t *off, int nq,   \
+      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
       int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
       unsigned *cand_p, int fam_floats) {                                      \
-    ivf_scan_body<FAM, IS_IP>(q, cent, cb, sq_vmin, sq_scale, probes, keys,    \
-                              codes, off, nq, nprobe, d, m, dsub, k, stride,   \
-                              cand_d, cand_p, fam_floats);                     \
+    ivf_scan_body<FAM, IS_IP, REGSEL>(q, cent, cb, sq_vmin, sq_scale, probes,  \
+                                      keys, codes, off, nq, nprobe, d, m,      \
+                                      dsub, k, stride, cand_d, cand_p,         \
+                                      fam_floats);                             \
   }
 
-INSTANTIATE_SCAN(k_scan_pq_l2, 0, false)
-INSTANTIATE_SCAN(k_scan_pq_ip, 0, true)
-INSTANTIATE_SCAN(k_scan_ivfflat_l2, 1, false)
-INSTANTIATE_SCAN(k_scan_ivfflat_ip, 1, true)
-INSTANTIATE_SCAN(k_scan_sq8_l2, 2, false)
-INSTANTIATE_SCAN(k_scan_sq8_ip, 2, true)
-INSTANTIATE_SCAN(k_scan_sqf_l2, 3, false)
-INSTANTIATE_SCAN(k_scan_sqf_ip, 3, true)
+INSTANTIATE_SCAN(k_scan_pq_l2, 0, false, false)
+INSTANTIATE_SCAN(k_scan_pq_ip, 0, true, false)
+INSTANTIATE_SCAN(k_scan_ivfflat_l2, 1, false, false)
+INSTANTIATE_SCAN(k_scan_ivfflat_ip, 1, true, false)
+INSTANTIATE_SCAN(k_scan_sq8_l2, 2, false, false)
+INSTANTIATE_SCAN(k_scan_sq8_ip, 2, true, false)
+INSTANTIATE_SCAN(k_scan_sqf_l2, 3, false, false)
+INSTANTIATE_SCAN(k_scan_sqf_ip, 3, true, false)
+INSTANTIATE_SCAN(k_scan_pq_l2_rk, 0, false, true)
+INSTANTIATE_SCAN(k_scan_pq_ip_rk, 0, true, true)
+INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
+INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)
+INSTANTIATE_SCAN(k_scan_sq8_l2_rk, 2, false, true)
+INSTANTIATE_SCAN(k_scan_sq8_ip_rk, 2, true, true)
+INSTANTIATE_SCAN(k_scan_sqf_l2_rk, 3, false, true)
+INSTANTIATE_SCAN(k_scan_sqf_ip_rk, 3, true, true)
 
 // ---------------------------------------------------------------------------
 // merge scan candidates -> final (D, I) per query.
@@ -597,6 +737,68 @@ extern "C" __global__ __launch_bounds__(256) void k_merge_shards(
     int sh = slot / k, jj = slot % k;
     Dout[qi * k + j] = s.d[j];  // negated for maximize — quirk 2 kept
     Iout[qi * k + j] = ((long long)sh * nq + qi) * k + jj;  // global slot
+  }
+}
+
+// register-path candidate merge (k <= 16)
+extern "C" __global__ __launch_bounds__(256) void k_merge_cand_rk(
+    const float *__restrict__ cand_d, const unsigned *__restrict__ cand_p,
+    long long nq, int C, int k, const int64_t *__restrict__ ids, int is_ip,
+    float *__restrict__ D, int64_t *__restrict__ I) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  long long qi = blockIdx.x;
+  if (qi >= nq) return;
+  RegTopK<16> loc;
+  loc.init();
+  const float *cd = cand_d + qi * C;
+  const unsigned *cp = cand_p + qi * C;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    unsigned pos = cp[c];
+    if (pos != PAD_POS) {
+      unsigned idu = ids ? (unsigned)ids[pos] : pos;
+      loc.push(cd[c], idu);
+    }
+  }
+  __syncthreads();
+  float *od = reinterpret_cast<float *>(smem + REGSEL_LDS_BYTES);
+  unsigned *op = reinterpret_cast<unsigned *>(smem + REGSEL_LDS_BYTES + 16 * 4);
+  regtopk_block_extract<16>(loc, k, smem, od, op);
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    bool v = op[j] != PAD_POS;
+    if (v) {
+      D[qi * k + j] = is_ip ? -od[j] : od[j];
+      I[qi * k + j] = (long long)op[j];
+    } else {
+      D[qi * k + j] = is_ip ? -DFANN_FLT_MAX : DFANN_FLT_MAX;
+      I[qi * k + j] = -1;
+    }
+  }
+}
+
+// register-path shard merge (k <= 16)
+extern "C" __global__ __launch_bounds__(256) void k_merge_shards_rk(
+    const float *__restrict__ Dall, long long nq, int S, int k, int maximize,
+    float *__restrict__ Dout, int64_t *__restrict__ Iout) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  long long qi = blockIdx.x;
+  if (qi >= nq) return;
+  RegTopK<16> loc;
+  loc.init();
+  int C = S * k;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    int sh = c / k, j = c % k;
+    float v = Dall[((long long)sh * nq + qi) * k + j];
+    loc.push(maximize ? -v : v, (unsigned)(sh * k + j));
+  }
+  __syncthreads();
+  float *od = reinterpret_cast<float *>(smem + REGSEL_LDS_BYTES);
+  unsigned *op = reinterpret_cast<unsigned *>(smem + REGSEL_LDS_BYTES + 16 * 4);
+  regtopk_block_extract<16>(loc, k, smem, od, op);
+  for (int j = threadIdx.x; j < k; j += blockDim.x) {
+    unsigned slot = op[j];
+    int sh = slot / k, jj = slot % k;
+    Dout[qi * k + j] = od[j];
+    Iout[qi * k + j] = ((long long)sh * nq + qi) * k + jj;
   }
 }
 

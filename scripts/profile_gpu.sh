@@ -19,13 +19,13 @@ cd /tmp && export TMPDIR=/tmp
 
 BENCH_ARGS="${BENCH_ARGS:---steps 5 --warmup 2 --cpu-baseline 0}"
 
-rocprofv3 --kernel-trace --stats -d "$OUT/trace" -- \
+rocprofv3 --kernel-trace --stats --output-format csv -d "$OUT/trace" -- \
     python "$ROOT/bench.py" $BENCH_ARGS > "$OUT/bench_trace.json" 2> "$OUT/trace.log"
 
-rocprofv3 --pmc FETCH_SIZE -d "$OUT/pmc_fetch" -- \
+rocprofv3 --pmc FETCH_SIZE --output-format csv -d "$OUT/pmc_fetch" -- \
     python "$ROOT/bench.py" $BENCH_ARGS > /dev/null 2> "$OUT/pmc_fetch.log" || true
 
-rocprofv3 --pmc WRITE_SIZE -d "$OUT/pmc_write" -- \
+rocprofv3 --pmc WRITE_SIZE --output-format csv -d "$OUT/pmc_write" -- \
     python "$ROOT/bench.py" $BENCH_ARGS > /dev/null 2> "$OUT/pmc_write.log" || true
 
 echo "profile artifacts in $OUT"
