@@ -52,6 +52,18 @@ WORKLOADS = {
         metric=0, nq=10_000, k=10, centers=10_000, sigma=0.15,
         fixed_nprobe=16,
     ),
+    # BASELINE.json configs[4]-shaped at single-GPU scale (8-bit SQ,
+    # HBM-stress flavor: 128 code bytes/vector)
+    "ivfsq8_1m_d128": dict(
+        type="ivfsq", d=128, n=1_000_000, nlist=1024, m=0, nbits=8,
+        metric=1, nq=10_000, k=10, centers=10_000, sigma=0.5,
+        sq_type="8bit",
+    ),
+    # 10M single-GPU stress (toward configs[3] scale)
+    "ivfpq_10m_d128_m16": dict(
+        type="ivfpq", d=128, n=10_000_000, nlist=4096, m=16, nbits=8,
+        metric=1, nq=10_000, k=10, centers=50_000, sigma=0.5,
+    ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
         type="ivfpq", d=64, n=100_000, nlist=256, m=8, nbits=8,
@@ -155,7 +167,7 @@ def main():
 
     spec = {"type": cfg["type"], "dim": cfg["d"], "metric": metric,
             "nlist": cfg["nlist"], "m": cfg["m"], "nbits": cfg["nbits"],
-            "nprobe": 1, "seed": 1234}
+            "sq_type": cfg.get("sq_type", "fp16"), "nprobe": 1, "seed": 1234}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)
@@ -246,7 +258,8 @@ def main():
         "unit": "GB/s",
         "frac": scan_gbs / HBM_PEAK_GBS,
         "traffic": None,  # PMC pass: see profiles/ (rocprofv3 --pmc)
-        "kernel": "k_scan_pq_l2" if cfg["type"] == "ivfpq" else "k_scan_ivfflat",
+        "kernel": {"ivfpq": "k_scan_pq", "ivfsq": "k_scan_sq8",
+                   "ivf_flat": "k_scan_ivfflat"}[cfg["type"]],
         "per_launch_bytes": per_launch_bytes,
         "per_launch_ms": per_launch_ms,
         "gemm_ms_frac": t["gemm_ms"] / max(t["scan_ms"] + t["gemm_ms"]
@@ -264,14 +277,20 @@ def main():
         orc.centroids = eng.get_centroids()
         if cfg["type"] == "ivfpq":
             orc.codebooks = eng.get_codebooks()
+        if cfg["type"] == "ivfsq" and cfg.get("sq_type") == "8bit":
+            vmin, vdiff = eng.get_sq_params()
+            orc.vmin, orc.vdiff = vmin, vdiff
+            orc.scale = (vdiff / np.float32(255.0)).astype(np.float32)
         orc.is_trained = True
         off, ids, codes = eng.get_lists()
-        cb = cfg["m"] if cfg["type"] == "ivfpq" else None
         for L in range(cfg["nlist"]):
             s0, s1 = int(off[L]), int(off[L + 1])
             orc.list_ids[L] = ids[s0:s1]
             if cfg["type"] == "ivfpq":
                 orc.list_codes[L] = codes[s0:s1, :cfg["m"]]
+            elif cfg["type"] == "ivfsq":
+                nb = cfg["d"] if cfg.get("sq_type") == "8bit" else 2 * cfg["d"]
+                orc.list_codes[L] = codes[s0:s1, :nb]
             else:
                 orc.list_data[L] = (
                     codes[s0:s1].reshape(s1 - s0, -1)[:, :cfg["d"] * 4]
