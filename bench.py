@@ -59,10 +59,16 @@ WORKLOADS = {
         metric=1, nq=10_000, k=10, centers=10_000, sigma=0.5,
         sq_type="8bit",
     ),
-    # 10M single-GPU stress (toward configs[3] scale)
+    # 10M single-GPU stress (toward configs[3] scale; 160 MB codes)
     "ivfpq_10m_d128_m16": dict(
         type="ivfpq", d=128, n=10_000_000, nlist=4096, m=16, nbits=8,
         metric=1, nq=10_000, k=10, centers=50_000, sigma=0.5,
+    ),
+    # true HBM-bound scan: 10M x 128B SQ8 codes = 1.28 GB >> 256 MB L3
+    "ivfsq8_10m_d128": dict(
+        type="ivfsq", d=128, n=10_000_000, nlist=4096, m=0, nbits=8,
+        metric=1, nq=10_000, k=10, centers=50_000, sigma=0.5,
+        sq_type="8bit",
     ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
@@ -251,8 +257,16 @@ def main():
     scan_gbs = (t["scan_bytes"] / 1e9) / (t["scan_ms"] / 1e3) if t["scan_ms"] else 0.0
     per_launch_bytes = t["scan_bytes"] / max(t["scan_launches"], 1)
     per_launch_ms = t["scan_ms"] / max(t["scan_launches"], 1)
+    # honesty about the cache level that actually bounds the scan: if the
+    # whole code arena fits the 256 MiB Infinity Cache, re-reads are
+    # absorbed on-die and the 8 TB/s HBM peak is not the binding roof
+    # (MI355X_MICROARCH.md §Infinity Cache)
+    code_bytes = {"ivfpq": cfg["m"], "ivfsq": cfg["d"],
+                  "ivf_flat": 4 * cfg["d"]}[cfg["type"]]
+    codes_total = eng.ntotal * ((code_bytes + 15) // 16 * 16)
+    hbm_bound = codes_total > 256 * 1024 * 1024
     roofline = {
-        "bound": "hbm",
+        "bound": "hbm" if hbm_bound else "l3-resident (codes fit 256MiB LLC)",
         "achieved": scan_gbs,
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
