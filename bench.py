@@ -158,12 +158,10 @@ def main():
     if world > 1:
         import torch.distributed as dist
 
-        xb0 = xb if rank == 0 else torch.empty_like(xb)
-        # avoid broadcasting 512MB: only the selected query rows
+        # queries derive from rank 0's shard; broadcast only the selected rows
         qrows = xb[qidx] if rank == 0 else torch.empty(
             cfg["nq"], cfg["d"], device=device)
         dist.broadcast(qrows, src=0)
-        del xb0
     else:
         qrows = xb[qidx]
     gq = torch.Generator(device=device).manual_seed(9999 + 7)

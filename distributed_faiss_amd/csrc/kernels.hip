@@ -169,22 +169,21 @@ struct RegTopK {
   }
 };
 
-// Block-wide merge of per-thread RegTopK heads: k extraction rounds.
-// lds: 48 bytes of scratch. out_d/out_p: k contiguous results (ascending).
-#define REGSEL_LDS_BYTES 64
+// Block-wide merge of per-thread RegTopK heads. Phase 1: each wave
+// extracts its own top-k with pure shfl rounds (no barriers); phase 2:
+// one barrier, then thread 0 4-way-merges the per-wave sorted lists.
+// lds: 4*16*8 = 512 B scratch. Requires blockDim == 256 (4 waves).
+#define REGSEL_LDS_BYTES 640
 template <int K>
 __device__ void regtopk_block_extract(RegTopK<K> &loc, int k, char *lds,
                                       float *out_d, unsigned *out_p) {
-  float *wd = reinterpret_cast<float *>(lds);           // per-wave best dist
-  unsigned *wp = reinterpret_cast<unsigned *>(lds + 16); // per-wave best pos
-  unsigned *wl = reinterpret_cast<unsigned *>(lds + 32); // per-wave best lane
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  int nw = blockDim.x >> 6;
+  float *wvd = reinterpret_cast<float *>(lds);             // [4][K]
+  unsigned *wvp = reinterpret_cast<unsigned *>(lds + 4 * K * 4);
   for (int round = 0; round < k; ++round) {
     float cd = loc.d[0];
     unsigned cp = loc.p[0];
-    unsigned cl = threadIdx.x;
-    // wave reduction (keep the smaller (d,p))
+    unsigned cl = (unsigned)lane;
 #pragma unroll
     for (int o = 32; o > 0; o >>= 1) {
       float od = __shfl_down(cd, o, 64);
@@ -192,25 +191,43 @@ __device__ void regtopk_block_extract(RegTopK<K> &loc, int k, char *lds,
       unsigned ol = __shfl_down(cl, o, 64);
       if (sel_less(od, op, cd, cp)) { cd = od; cp = op; cl = ol; }
     }
-    if (lane == 0) { wd[w] = cd; wp[w] = cp; wl[w] = cl; }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      float bd = wd[0]; unsigned bp = wp[0], bl = wl[0];
-      for (int i = 1; i < nw; ++i)
-        if (sel_less(wd[i], wp[i], bd, bp)) { bd = wd[i]; bp = wp[i]; bl = wl[i]; }
-      wd[0] = bd; wp[0] = bp; wl[0] = bl;
+    cd = __shfl(cd, 0, 64);
+    cp = __shfl(cp, 0, 64);
+    cl = __shfl(cl, 0, 64);
+    if (lane == 0) {
+      wvd[w * K + round] = cd;
+      wvp[w * K + round] = cp;
     }
-    __syncthreads();
-    float bd = wd[0];
-    unsigned bp = wp[0], bl = wl[0];
-    if (threadIdx.x == 0) {
-      bool valid = bp != PAD_POS;
-      out_d[round] = valid ? bd : DFANN_FLT_MAX;
-      out_p[round] = valid ? bp : PAD_POS;
-    }
-    if (threadIdx.x == bl) loc.pop_head();
-    __syncthreads();
+    if ((unsigned)lane == cl) loc.pop_head();
   }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int c0 = 0, c1 = 0, c2 = 0, c3 = 0;
+    for (int round = 0; round < k; ++round) {
+      // a wave list holds exactly k entries; an exhausted list reads as pad
+      float b0 = c0 < k ? wvd[0 * K + c0] : DFANN_FLT_MAX;
+      float b1 = c1 < k ? wvd[1 * K + c1] : DFANN_FLT_MAX;
+      float b2 = c2 < k ? wvd[2 * K + c2] : DFANN_FLT_MAX;
+      float b3 = c3 < k ? wvd[3 * K + c3] : DFANN_FLT_MAX;
+      unsigned p0 = c0 < k ? wvp[0 * K + c0] : PAD_POS;
+      unsigned p1 = c1 < k ? wvp[1 * K + c1] : PAD_POS;
+      unsigned p2 = c2 < k ? wvp[2 * K + c2] : PAD_POS;
+      unsigned p3 = c3 < k ? wvp[3 * K + c3] : PAD_POS;
+      int bw = 0;
+      float bd = b0;
+      unsigned bp = p0;
+      if (sel_less(b1, p1, bd, bp)) { bd = b1; bp = p1; bw = 1; }
+      if (sel_less(b2, p2, bd, bp)) { bd = b2; bp = p2; bw = 2; }
+      if (sel_less(b3, p3, bd, bp)) { bd = b3; bp = p3; bw = 3; }
+      out_d[round] = bd;
+      out_p[round] = bp;
+      if (bw == 0) ++c0;
+      else if (bw == 1) ++c1;
+      else if (bw == 2) ++c2;
+      else ++c3;
+    }
+  }
+  __syncthreads();
 }
 
 // ---------------------------------------------------------------------------
@@ -415,66 +432,103 @@ extern "C" __global__ void k_assign_chunk(const float *__restrict__ keys,
 
 // per-row distance (shared by both selection paths); fam = staged LDS
 // region (FAM 0: LUT; FAM 2: [target][vmin][scale]; FAM 3: target).
+// Loads are issued in 64-byte batches (4x uint4 up front) so the HBM
+// latency of a row's chunks overlaps instead of chaining — the
+// accumulation ORDER is unchanged (sequential over the reduced axis,
+// contract off), so the oracle bit-exactness contract holds.
+
+#define DFANN_PROC16_PQ(WV, G)                                                 \
+  if ((G) < m) {                                                               \
+    unsigned w0_ = (WV).x, w1_ = (WV).y, w2_ = (WV).z, w3_ = (WV).w;           \
+    _Pragma("unroll") for (int b = 0; b < 16; ++b) {                           \
+      if ((G) + b < m) {                                                       \
+        unsigned word = (b < 4) ? w0_ : (b < 8) ? w1_ : (b < 12) ? w2_ : w3_;  \
+        unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;                          \
+        acc = acc + lut[((G) + b) * 256 + c];                                  \
+      }                                                                        \
+    }                                                                          \
+  }
+
+#define DFANN_PROC16_SQ8(WV, G)                                                \
+  if ((G) < d) {                                                               \
+    unsigned w0_ = (WV).x, w1_ = (WV).y, w2_ = (WV).z, w3_ = (WV).w;           \
+    _Pragma("unroll") for (int b = 0; b < 16; ++b) {                           \
+      if ((G) + b < d) {                                                       \
+        unsigned word = (b < 4) ? w0_ : (b < 8) ? w1_ : (b < 12) ? w2_ : w3_;  \
+        unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;                          \
+        int t = (G) + b;                                                       \
+        float dec = vm[t] + ((float)c + 0.5f) * sc[t];                         \
+        if (IS_IP) acc = acc + rbuf[t] * dec;                                  \
+        else {                                                                 \
+          float diff = rbuf[t] - dec;                                          \
+          acc = acc + diff * diff;                                             \
+        }                                                                      \
+      }                                                                        \
+    }                                                                          \
+  }
+
+#define DFANN_PROC8_F16(WV, G)                                                 \
+  if ((G) < d) {                                                               \
+    _Pragma("unroll") for (int b = 0; b < 8; ++b) {                            \
+      if ((G) + b < d) {                                                       \
+        unsigned word = (b < 2) ? (WV).x : (b < 4) ? (WV).y                    \
+                                 : (b < 6) ? (WV).z : (WV).w;                  \
+        unsigned hv = (word >> ((b & 1) * 16)) & 0xFFFFu;                      \
+        float dec = __half2float(__ushort_as_half((unsigned short)hv));        \
+        int t = (G) + b;                                                       \
+        if (IS_IP) acc = acc + rbuf[t] * dec;                                  \
+        else {                                                                 \
+          float diff = rbuf[t] - dec;                                          \
+          acc = acc + diff * diff;                                             \
+        }                                                                      \
+      }                                                                        \
+    }                                                                          \
+  }
+
 template <int FAM, bool IS_IP>
 __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
                                                const float *__restrict__ fam,
                                                int d, int m) {
   float acc = 0.f;
+  const uint4 zero4 = {0, 0, 0, 0};
   if (FAM == 0) {
     const float *lut = fam;
-    for (int g = 0; g < m; g += 16) {
+    for (int g0 = 0; g0 < m; g0 += 64) {
 #pragma clang fp contract(off)
-      uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
-      unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
-#pragma unroll
-      for (int b = 0; b < 16; ++b) {
-        if (g + b < m) {
-          unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
-          unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
-          acc = acc + lut[(g + b) * 256 + c];
-        }
-      }
+      uint4 wa = *reinterpret_cast<const uint4 *>(cp + g0);
+      uint4 wb = (g0 + 16 < m) ? *reinterpret_cast<const uint4 *>(cp + g0 + 16) : zero4;
+      uint4 wc = (g0 + 32 < m) ? *reinterpret_cast<const uint4 *>(cp + g0 + 32) : zero4;
+      uint4 wd4 = (g0 + 48 < m) ? *reinterpret_cast<const uint4 *>(cp + g0 + 48) : zero4;
+      DFANN_PROC16_PQ(wa, g0)
+      DFANN_PROC16_PQ(wb, g0 + 16)
+      DFANN_PROC16_PQ(wc, g0 + 32)
+      DFANN_PROC16_PQ(wd4, g0 + 48)
     }
   } else if (FAM == 2) {
     const float *rbuf = fam, *vm = fam + d, *sc = fam + 2 * d;
-    for (int g = 0; g < d; g += 16) {
+    for (int g0 = 0; g0 < d; g0 += 64) {
 #pragma clang fp contract(off)
-      uint4 wv = *reinterpret_cast<const uint4 *>(cp + g);
-      unsigned w0 = wv.x, w1 = wv.y, w2 = wv.z, w3 = wv.w;
-#pragma unroll
-      for (int b = 0; b < 16; ++b) {
-        if (g + b < d) {
-          unsigned word = (b < 4) ? w0 : (b < 8) ? w1 : (b < 12) ? w2 : w3;
-          unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;
-          int t = g + b;
-          float dec = vm[t] + ((float)c + 0.5f) * sc[t];
-          if (IS_IP) acc = acc + rbuf[t] * dec;
-          else {
-            float diff = rbuf[t] - dec;
-            acc = acc + diff * diff;
-          }
-        }
-      }
+      uint4 wa = *reinterpret_cast<const uint4 *>(cp + g0);
+      uint4 wb = (g0 + 16 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 16) : zero4;
+      uint4 wc = (g0 + 32 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 32) : zero4;
+      uint4 wd4 = (g0 + 48 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 48) : zero4;
+      DFANN_PROC16_SQ8(wa, g0)
+      DFANN_PROC16_SQ8(wb, g0 + 16)
+      DFANN_PROC16_SQ8(wc, g0 + 32)
+      DFANN_PROC16_SQ8(wd4, g0 + 48)
     }
-  } else {  // FAM 3: fp16 codes
+  } else {  // FAM 3: fp16 codes, 8 dims per 16 B
     const float *rbuf = fam;
-    for (int g = 0; g < d; g += 8) {
+    for (int g0 = 0; g0 < d; g0 += 32) {
 #pragma clang fp contract(off)
-      uint4 wv = *reinterpret_cast<const uint4 *>(cp + (size_t)g * 2);
-#pragma unroll
-      for (int b = 0; b < 8; ++b) {
-        if (g + b < d) {
-          unsigned word = (b < 2) ? wv.x : (b < 4) ? wv.y : (b < 6) ? wv.z : wv.w;
-          unsigned h = (word >> ((b & 1) * 16)) & 0xFFFFu;
-          float dec = __half2float(__ushort_as_half((unsigned short)h));
-          int t = g + b;
-          if (IS_IP) acc = acc + rbuf[t] * dec;
-          else {
-            float diff = rbuf[t] - dec;
-            acc = acc + diff * diff;
-          }
-        }
-      }
+      uint4 wa = *reinterpret_cast<const uint4 *>(cp + (size_t)g0 * 2);
+      uint4 wb = (g0 + 8 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 8) * 2) : zero4;
+      uint4 wc = (g0 + 16 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 16) * 2) : zero4;
+      uint4 wd4 = (g0 + 24 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 24) * 2) : zero4;
+      DFANN_PROC8_F16(wa, g0)
+      DFANN_PROC8_F16(wb, g0 + 8)
+      DFANN_PROC8_F16(wc, g0 + 16)
+      DFANN_PROC8_F16(wd4, g0 + 24)
     }
   }
   return acc;
@@ -587,9 +641,24 @@ __device__ void ivf_scan_body(
         }
       }
     }
+  } else if (REGSEL) {
+    // 4 independent rows per thread per iteration: their load batches
+    // overlap, hiding HBM/L2 latency without any block synchronization
+    for (long long base = s0; base < s1; base += 1024) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        long long pos = base + (long long)u * 256 + threadIdx.x;
+        if (pos < s1) {
+          const uint8_t *cp = codes + pos * (size_t)stride;
+          float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
+          float dist = IS_IP ? -(bias + acc) : acc;
+          loc.push(dist, (unsigned)pos);
+        }
+      }
+    }
   } else {
     for (long long base = s0; base < s1; base += 512) {
-      if (!REGSEL) sel_guard(s, k, 512);
+      sel_guard(s, k, 512);
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
         long long pos = base + (long long)u * 256 + threadIdx.x;
@@ -597,8 +666,7 @@ __device__ void ivf_scan_body(
           const uint8_t *cp = codes + pos * (size_t)stride;
           float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
           float dist = IS_IP ? -(bias + acc) : acc;
-          if (REGSEL) loc.push(dist, (unsigned)pos);
-          else sel_try(s, dist, (unsigned)pos);
+          sel_try(s, dist, (unsigned)pos);
         }
       }
     }
