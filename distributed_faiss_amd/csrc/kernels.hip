@@ -505,30 +505,50 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
       DFANN_PROC16_PQ(wd4, g0 + 48)
     }
   } else if (FAM == 2) {
+    // full-cacheline batches: a row is one 128-B line (stride 128 for
+    // d=128); reading it in one 8x-uint4 burst touches the line once —
+    // split batches re-fetched it after L1 eviction (229 KB of rows in
+    // flight per CU >> 32 KB L1)
     const float *rbuf = fam, *vm = fam + d, *sc = fam + 2 * d;
-    for (int g0 = 0; g0 < d; g0 += 64) {
+    for (int g0 = 0; g0 < d; g0 += 128) {
 #pragma clang fp contract(off)
       uint4 wa = *reinterpret_cast<const uint4 *>(cp + g0);
       uint4 wb = (g0 + 16 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 16) : zero4;
       uint4 wc = (g0 + 32 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 32) : zero4;
       uint4 wd4 = (g0 + 48 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 48) : zero4;
+      uint4 we = (g0 + 64 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 64) : zero4;
+      uint4 wf = (g0 + 80 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 80) : zero4;
+      uint4 wg = (g0 + 96 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 96) : zero4;
+      uint4 wh = (g0 + 112 < d) ? *reinterpret_cast<const uint4 *>(cp + g0 + 112) : zero4;
       DFANN_PROC16_SQ8(wa, g0)
       DFANN_PROC16_SQ8(wb, g0 + 16)
       DFANN_PROC16_SQ8(wc, g0 + 32)
       DFANN_PROC16_SQ8(wd4, g0 + 48)
+      DFANN_PROC16_SQ8(we, g0 + 64)
+      DFANN_PROC16_SQ8(wf, g0 + 80)
+      DFANN_PROC16_SQ8(wg, g0 + 96)
+      DFANN_PROC16_SQ8(wh, g0 + 112)
     }
   } else {  // FAM 3: fp16 codes, 8 dims per 16 B
     const float *rbuf = fam;
-    for (int g0 = 0; g0 < d; g0 += 32) {
+    for (int g0 = 0; g0 < d; g0 += 64) {  // 128 B = one line per batch
 #pragma clang fp contract(off)
       uint4 wa = *reinterpret_cast<const uint4 *>(cp + (size_t)g0 * 2);
       uint4 wb = (g0 + 8 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 8) * 2) : zero4;
       uint4 wc = (g0 + 16 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 16) * 2) : zero4;
       uint4 wd4 = (g0 + 24 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 24) * 2) : zero4;
+      uint4 we = (g0 + 32 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 32) * 2) : zero4;
+      uint4 wf = (g0 + 40 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 40) * 2) : zero4;
+      uint4 wg = (g0 + 48 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 48) * 2) : zero4;
+      uint4 wh = (g0 + 56 < d) ? *reinterpret_cast<const uint4 *>(cp + (size_t)(g0 + 56) * 2) : zero4;
       DFANN_PROC8_F16(wa, g0)
       DFANN_PROC8_F16(wb, g0 + 8)
       DFANN_PROC8_F16(wc, g0 + 16)
       DFANN_PROC8_F16(wd4, g0 + 24)
+      DFANN_PROC8_F16(we, g0 + 32)
+      DFANN_PROC8_F16(wf, g0 + 40)
+      DFANN_PROC8_F16(wg, g0 + 48)
+      DFANN_PROC8_F16(wh, g0 + 56)
     }
   }
   return acc;
@@ -642,11 +662,16 @@ __device__ void ivf_scan_body(
       }
     }
   } else if (REGSEL) {
-    // 4 independent rows per thread per iteration: their load batches
-    // overlap, hiding HBM/L2 latency without any block synchronization
-    for (long long base = s0; base < s1; base += 1024) {
+    // independent rows per thread per iteration: their load batches
+    // overlap, hiding HBM/L2 latency without any block synchronization.
+    // FAM 0 rows are 16-64 B (4 in flight, cheap); FAM 2/3 rows are a
+    // full cache line each (8 uint4 in registers), so 2 in flight keeps
+    // VGPRs ~100 (5 waves/SIMD) instead of 161 (3 waves).
+    const int UROWS = (FAM == 0) ? 4 : 2;
+    for (long long base = s0; base < s1; base += (long long)UROWS * 256) {
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
+        if (u >= UROWS) break;
         long long pos = base + (long long)u * 256 + threadIdx.x;
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
