@@ -663,7 +663,7 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
       break;
     case T_IVFSQ:
       if (h->sq8) {
-        fam_floats = 3 * h->d;
+        fam_floats = 2 * h->d;
         kern = rk ? (ip ? k_scan_sq8_ip_rk : k_scan_sq8_l2_rk)
                   : (ip ? k_scan_sq8_ip : k_scan_sq8_l2);
       } else {

@@ -293,6 +293,88 @@ extern "C" __global__ __launch_bounds__(256) void k_gemm_nt(
 }
 
 // ---------------------------------------------------------------------------
+// bf16 GEMM: C[i][j] = sum_k A[i][k]*B[j][k], A/B bf16 (pre-converted),
+// C fp32. v_mfma_f32_16x16x32_bf16, 128x128 tile, 4 waves x (4x4 of
+// 16x16). Used for the APPROXIMATE assign/coarse path on huge nlist
+// (spec "coarse_bf16") — ~16x the f32 MFMA rate.
+// Fragment maps (verified on hardware by tests/test_gpu_parity.py):
+//   A: lane l supplies A[i = l&15][k = (l>>4)*8 + e], e in 0..7
+//   B: lane l supplies B[k = (l>>4)*8 + e][j = l&15]
+//   C/D: col = lane&15, row = (lane>>4)*4 + reg, reg in 0..3
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+
+extern "C" __global__ void k_f32_to_bf16(const float *__restrict__ in,
+                                         long long n,
+                                         unsigned short *__restrict__ out) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i < n; i += (long long)gridDim.x * blockDim.x) {
+    union { float f; unsigned u; } v;
+    v.f = in[i];
+    // round-to-nearest-even bf16 truncation
+    unsigned lsb = (v.u >> 16) & 1u;
+    out[i] = (unsigned short)((v.u + 0x7FFFu + lsb) >> 16);
+  }
+}
+
+#define GB_T 128
+#define GB_K 32
+
+extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
+    const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc) {
+  __shared__ unsigned short sA[GB_T][GB_K + 8];
+  __shared__ unsigned short sB[GB_T][GB_K + 8];
+  int bi = blockIdx.y * GB_T;
+  int bj = blockIdx.x * GB_T;
+  int tid = threadIdx.x;
+  int lane = tid & 63, w = tid >> 6;
+  // wave w covers rows [wr, wr+64) x cols [wc, wc+64): 4x4 tiles of 16x16
+  int wr = (w >> 1) * 64, wc = (w & 1) * 64;
+  f32x4 acc[4][4] = {};
+  int li = lane & 15;           // fragment row/col within tile
+  int ke = (lane >> 4) * 8;     // k-offset of this lane's 8 elements
+  for (int k0 = 0; k0 < K; k0 += GB_K) {
+    for (int e = tid; e < GB_T * GB_K; e += 256) {
+      int r = e >> 5, c = e & 31;
+      sA[r][c] = (bi + r < M && k0 + c < K)
+                     ? A[(size_t)(bi + r) * lda + k0 + c] : (unsigned short)0;
+      sB[r][c] = (bj + r < N && k0 + c < K)
+                     ? B[(size_t)(bj + r) * ldb + k0 + c] : (unsigned short)0;
+    }
+    __syncthreads();
+    // one mfma_f32_16x16x32_bf16 covers a full K=32 slice across the wave
+    // (lane l holds k = (l>>4)*8 + 0..7), so GB_K == 32 needs exactly one
+    // MFMA per (ti,tj)
+#pragma unroll
+    for (int ti = 0; ti < 4; ++ti) {
+      bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&sA[wr + ti * 16 + li][ke]);
+#pragma unroll
+      for (int tj = 0; tj < 4; ++tj) {
+        bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(&sB[wc + tj * 16 + li][ke]);
+        acc[ti][tj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+  int rrow = (lane >> 4) * 4;
+#pragma unroll
+  for (int ti = 0; ti < 4; ++ti) {
+#pragma unroll
+    for (int tj = 0; tj < 4; ++tj) {
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        int row = bi + wr + ti * 16 + rrow + rg;
+        int col = bj + wc + tj * 16 + li;
+        if (row < M && col < N) C[(size_t)row * ldc + col] = acc[ti][tj][rg];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // row norms ||x_i||^2, one block per row
 // ---------------------------------------------------------------------------
 
@@ -449,18 +531,27 @@ extern "C" __global__ void k_assign_chunk(const float *__restrict__ keys,
     }                                                                          \
   }
 
+// SQ8 distance with the decode algebra FOLDED:
+// L2:  diff = u[t] - c*v[t],  u = (q-cent-vmin) - 0.5*scale, v = scale
+// IP:  acc += u[t] + c*v[t],  u = q*vmin + 0.5*(q*scale),   v = q*scale
+// — mathematically the faiss codec, op-for-op mirrored by the oracle
+// (oracle/core.py OracleIVFSQ._scan_one) so distances stay bitwise equal.
+// v_cvt_f32_ubyteN converts the byte exactly (one instruction).
+// (float)((w >> 8n) & 0xff) — LLVM selects v_cvt_f32_ubyteN for this
+#define DFANN_CVT_UB(WORD, B) ((float)(((WORD) >> (((B) & 3) * 8)) & 0xFFu))
+
 #define DFANN_PROC16_SQ8(WV, G)                                                \
   if ((G) < d) {                                                               \
     unsigned w0_ = (WV).x, w1_ = (WV).y, w2_ = (WV).z, w3_ = (WV).w;           \
     _Pragma("unroll") for (int b = 0; b < 16; ++b) {                           \
       if ((G) + b < d) {                                                       \
         unsigned word = (b < 4) ? w0_ : (b < 8) ? w1_ : (b < 12) ? w2_ : w3_;  \
-        unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;                          \
+        float cf = DFANN_CVT_UB(word, b);                                      \
         int t = (G) + b;                                                       \
-        float dec = vm[t] + ((float)c + 0.5f) * sc[t];                         \
-        if (IS_IP) acc = acc + rbuf[t] * dec;                                  \
-        else {                                                                 \
-          float diff = rbuf[t] - dec;                                          \
+        if (IS_IP) {                                                           \
+          acc = acc + (ubuf[t] + cf * vbuf[t]);                                \
+        } else {                                                               \
+          float diff = ubuf[t] - cf * vbuf[t];                                 \
           acc = acc + diff * diff;                                             \
         }                                                                      \
       }                                                                        \
@@ -509,7 +600,7 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
     // d=128); reading it in one 8x-uint4 burst touches the line once —
     // split batches re-fetched it after L1 eviction (229 KB of rows in
     // flight per CU >> 32 KB L1)
-    const float *rbuf = fam, *vm = fam + d, *sc = fam + 2 * d;
+    const float *ubuf = fam, *vbuf = fam + d;
     for (int g0 = 0; g0 < d; g0 += 128) {
 #pragma clang fp contract(off)
       uint4 wa = *reinterpret_cast<const uint4 *>(cp + g0);
@@ -609,11 +700,20 @@ __device__ void ivf_scan_body(
   } else if (FAM == 1) {
     for (int t = threadIdx.x; t < d; t += blockDim.x) fam[t] = qp[t];
   } else if (FAM == 2) {
-    float *rbuf = fam, *vm = fam + d, *sc = fam + 2 * d;
+    // folded SQ8 terms (see DFANN_PROC16_SQ8); exact op order mirrored in
+    // the oracle: 0.5f*scale is an exact exponent decrement
+    float *ubuf = fam, *vbuf = fam + d;
     for (int t = threadIdx.x; t < d; t += blockDim.x) {
-      rbuf[t] = IS_IP ? qp[t] : qp[t] - cent[(long long)L * d + t];
-      vm[t] = sq_vmin[t];
-      sc[t] = sq_scale[t];
+      float sct = sq_scale[t];
+      if (IS_IP) {
+        float qsc = qp[t] * sct;
+        vbuf[t] = qsc;
+        ubuf[t] = qp[t] * sq_vmin[t] + 0.5f * qsc;
+      } else {
+        float r = qp[t] - cent[(long long)L * d + t];
+        ubuf[t] = (r - sq_vmin[t]) - 0.5f * sct;
+        vbuf[t] = sct;
+      }
     }
   } else {  // SQfp16
     for (int t = threadIdx.x; t < d; t += blockDim.x)

@@ -623,7 +623,32 @@ class OracleIVFSQ(_OracleIVFBase):
         codes = self.list_codes[li]
         if codes.shape[0] == 0:
             return None
-        dec = self._decode_codes(codes)  # residual values
+        if self.qtype == "8bit":
+            # folded decode algebra, op-for-op the HIP scan kernel
+            # (csrc/kernels.hip DFANN_PROC16_SQ8): L2 diff = u - c*v with
+            # u = (q-cent-vmin) - 0.5*scale, v = scale; IP acc += u + c*v
+            # with u = q*vmin + 0.5*(q*scale), v = q*scale. Equal to the
+            # faiss codec in exact arithmetic; this is the shared fp32
+            # rounding order of the bit-exact tier.
+            cf = codes.astype(np.float32)
+            if self.metric == METRIC_L2:
+                r = (qi - self.centroids[li]).astype(np.float32)
+                u = (r - self.vmin) - np.float32(0.5) * self.scale
+                v = self.scale
+                acc = np.zeros(codes.shape[0], dtype=np.float32)
+                for t in range(self.d):
+                    diff = u[t] - cf[:, t] * v[t]
+                    acc = acc + diff * diff
+                return acc, self.list_ids[li]
+            qsc = (qi * self.scale).astype(np.float32)
+            u = qi * self.vmin + np.float32(0.5) * qsc
+            acc = np.zeros(codes.shape[0], dtype=np.float32)
+            for t in range(self.d):
+                acc = acc + (u[t] + cf[:, t] * qsc[t])
+            if bias is None:
+                bias = np.float32(seq_ip(qi, self.centroids[li][None, :])[0])
+            return bias + acc, self.list_ids[li]
+        dec = self._decode_codes(codes)  # fp16: residual values
         if self.metric == METRIC_L2:
             r = (qi - self.centroids[li]).astype(np.float32)
             return seq_l2(r, dec), self.list_ids[li]
