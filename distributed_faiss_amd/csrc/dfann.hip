@@ -191,6 +191,9 @@ struct dfann_index {
   int code_bytes = 0, stride = 0;
   int ws_mb = 512;  // chunk budget for key matrices (spec "ws_mb"; tests
                     // shrink it to force the multi-chunk paths)
+  bool coarse_bf16 = false;  // spec "coarse_bf16": assign/coarse GEMMs on
+                             // bf16 MFMA (~16x f32 rate) — approximate
+                             // ranking path for huge nlist
 
   DevBuf centroids, cnorm, codebooks, sq_vmin, sq_vdiff, sq_scale;
   // staging (arrival order)
@@ -203,7 +206,8 @@ struct dfann_index {
   // flat arena
   DevBuf flat;
   // workspace
-  DevBuf ws1, ws2, ws3, ws4, ws5;
+  DevBuf ws1, ws2, ws3, ws4, ws5, ws_bf16a, ws_bf16b;
+  DevBuf cent_bf16;
 
   // timing
   bool timing = false;
@@ -246,21 +250,50 @@ static dim3 grid1d(int64_t total, int block = 256, int64_t cap = 65535LL * 8) {
 
 // C[i][j] keys for rows of A vs rows of B under `metric`.
 // mode: 0 -ip, 1 bn-2ip, 2 qn+bn-2ip (needs qn). Result in `keys` (rows x N).
-static void gemm_keys(dfann_index *h, const float *A, int64_t Mrows,
-                      const float *B, int64_t N, int K, const float *bn,
-                      const float *qn, int mode, float *keys,
-                      hipStream_t stream) {
+// B_bf16: pre-converted B for the approximate bf16 path (used only when
+// h->coarse_bf16 and the caller provides it).
+static void gemm_keys_b(dfann_index *h, const float *A, int64_t Mrows,
+                        const float *B, const unsigned short *B_bf16,
+                        int64_t N, int K, const float *bn, const float *qn,
+                        int mode, float *keys, hipStream_t stream) {
   dim3 g((unsigned)((N + GT - 1) / GT), (unsigned)((Mrows + GT - 1) / GT));
   TimingEv e;
   if (h && h->timing) e = h->ev_begin(stream);
-  hipLaunchKernelGGL(k_gemm_nt, g, dim3(256), 0, stream, A, B, keys, (int)Mrows,
-                     (int)N, K, K, K, (int)N);
+  if (h && h->coarse_bf16 && B_bf16) {
+    h->ws_bf16a.ensure((size_t)Mrows * K * 2);
+    hipLaunchKernelGGL(k_f32_to_bf16, grid1d(Mrows * K), dim3(256), 0, stream,
+                       A, Mrows * K, h->ws_bf16a.as<unsigned short>());
+    hipLaunchKernelGGL(k_gemm_bf16_nt, g, dim3(256), 0, stream,
+                       h->ws_bf16a.as<unsigned short>(), B_bf16, keys,
+                       (int)Mrows, (int)N, K, K, K, (int)N);
+  } else {
+    hipLaunchKernelGGL(k_gemm_nt, g, dim3(256), 0, stream, A, B, keys,
+                       (int)Mrows, (int)N, K, K, K, (int)N);
+  }
   hipLaunchKernelGGL(k_score_key, grid1d(Mrows * N), dim3(256), 0, stream, keys,
                      Mrows, N, N, qn, bn, mode);
   if (h && h->timing) {
     h->ev_end(e, stream, h->ev_gemm);
     h->gemm_flops += 2LL * Mrows * N * K;
   }
+  HIP_CHECK(hipGetLastError());
+}
+
+static void gemm_keys(dfann_index *h, const float *A, int64_t Mrows,
+                      const float *B, int64_t N, int K, const float *bn,
+                      const float *qn, int mode, float *keys,
+                      hipStream_t stream) {
+  gemm_keys_b(h, A, Mrows, B, nullptr, N, K, bn, qn, mode, keys, stream);
+}
+
+// refresh the bf16 centroid image after (re)training / loading
+static void refresh_cent_bf16(dfann_index *h, hipStream_t stream) {
+  if (!h->coarse_bf16 || !h->centroids.p) return;
+  h->cent_bf16.ensure((size_t)h->nlist * h->d * 2);
+  hipLaunchKernelGGL(k_f32_to_bf16, grid1d((int64_t)h->nlist * h->d), dim3(256),
+                     0, stream, h->centroids.as<float>(),
+                     (long long)h->nlist * h->d,
+                     h->cent_bf16.as<unsigned short>());
   HIP_CHECK(hipGetLastError());
 }
 
@@ -287,9 +320,10 @@ static void assign_rows(dfann_index *h, const float *x, int64_t n,
                      assign_dev, n);
   for (int64_t s = 0; s < n; s += chunk) {
     int64_t c = std::min(chunk, n - s);
-    gemm_keys(h, x + s * h->d, c, h->centroids.as<float>(), nlist, h->d,
-              h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, keys,
-              stream);
+    gemm_keys_b(h, x + s * h->d, c, h->centroids.as<float>(),
+                h->cent_bf16.as<unsigned short>(), nlist, h->d,
+                h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, keys,
+                stream);
     hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream, keys, c,
                        (long long)nlist, (long long)nlist, 0, bestv + s,
                        assign_dev + s);
@@ -346,14 +380,23 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
   std::vector<int> h_counts(kcent);
   std::vector<float> h_cent;
 
+  DevBuf cent_b16;
+  bool use_b16 = h && h->coarse_bf16;
+  if (use_b16) cent_b16.ensure((size_t)kcent * d * 2);
   for (int it = 0; it < NITER; ++it) {
     rownorms(cent_out, kcent, d, cn.as<float>(), stream);
+    if (use_b16)
+      hipLaunchKernelGGL(k_f32_to_bf16, grid1d((int64_t)kcent * d), dim3(256),
+                         0, stream, cent_out, (long long)kcent * d,
+                         cent_b16.as<unsigned short>());
     hipLaunchKernelGGL(k_assign_init, grid1d(nt), dim3(256), 0, stream,
                        bestv.as<float>(), asg.as<int>(), nt);
     for (int64_t s = 0; s < nt; s += chunk) {
       int64_t c = std::min(chunk, nt - s);
-      gemm_keys(nullptr, xt + s * d, c, cent_out, kcent, d, cn.as<float>(),
-                nullptr, metric == M_IP ? 0 : 1, keys.as<float>(), stream);
+      gemm_keys_b(h, xt + s * d, c, cent_out,
+                  use_b16 ? cent_b16.as<unsigned short>() : nullptr, kcent, d,
+                  cn.as<float>(), nullptr, metric == M_IP ? 0 : 1,
+                  keys.as<float>(), stream);
       hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream,
                          keys.as<float>(), c, (long long)kcent,
                          (long long)kcent, 0, bestv.as<float>() + s,
@@ -423,6 +466,7 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->sq8 = json_str(js, "sq_type", "fp16") == "8bit";
   h->ws_mb = (int)json_int(js, "ws_mb", 512);
   if (h->ws_mb < 1) h->ws_mb = 1;
+  h->coarse_bf16 = json_int(js, "coarse_bf16", 0) != 0;
   if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
   if (h->type != T_FLAT && h->nlist <= 0) {
     delete h;
@@ -460,6 +504,7 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
   trace_point("train:coarse-kmeans", stream);
   rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(),
            stream);
+  refresh_cent_bf16(h, stream);
   if (h->type == T_IVFPQ || (h->type == T_IVFSQ && h->sq8)) {
     DevBuf asg, resid;
     asg.ensure((size_t)n * 4);
@@ -614,9 +659,10 @@ static void coarse_impl(dfann_index *h, int64_t nq, const float *q, int nprobe,
   float *sc = h->ws1.as<float>();
   for (int64_t s = 0; s < nq; s += chunk) {
     int64_t c = std::min(chunk, nq - s);
-    gemm_keys(h, q + s * h->d, c, h->centroids.as<float>(), nlist, h->d,
-              h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, sc,
-              stream);
+    gemm_keys_b(h, q + s * h->d, c, h->centroids.as<float>(),
+                h->cent_bf16.as<unsigned short>(), nlist, h->d,
+                h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, sc,
+                stream);
     if (use_regsel(nprobe)) {
       hipLaunchKernelGGL(k_topk_rows_rk, dim3((unsigned)c), dim3(256),
                          REGSEL_LDS_BYTES, stream, sc, c, (long long)nlist,
@@ -961,6 +1007,7 @@ extern "C" int dfann_set_trained(dfann_index *h, const float *centroids_host,
   HIP_CHECK(hipMemcpy(h->centroids.p, centroids_host,
                       (size_t)h->nlist * h->d * 4, hipMemcpyHostToDevice));
   rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
+  refresh_cent_bf16(h, 0);
   if (h->type == T_IVFPQ) {
     if (!codebooks_host) throw std::runtime_error("codebooks required");
     h->codebooks.ensure((size_t)h->m * 256 * h->dsub * 4);
@@ -1103,6 +1150,7 @@ extern "C" int dfann_load(const char *path, dfann_index **out) {
       load_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
       h->cnorm.ensure((size_t)h->nlist * 4);
       rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
+      refresh_cent_bf16(h, 0);
       if (h->type == T_IVFPQ)
         load_dev(f, h->codebooks, (size_t)h->m * 256 * h->dsub * 4);
       if (h->type == T_IVFSQ && h->sq8) {

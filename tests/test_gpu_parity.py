@@ -233,6 +233,41 @@ def test_ivfflat_tolerance_parity():
     np.testing.assert_allclose(D[same], Do[same], rtol=1e-4, atol=1e-4)
 
 
+@pytest.mark.parametrize("metric", [IP, L2])
+def test_coarse_bf16_path(metric):
+    # approximate bf16 assign/coarse (spec coarse_bf16): on well-separated
+    # clusters the probe sets match fp32 exactly, so end-to-end results are
+    # identical — and a wrong MFMA fragment layout would wreck every probe
+    d, nlist = 64, 16
+    cent, xb = _clustered(nlist, 300, d, seed=21)
+    q = xb[::11][:30] + 0.01 * _rand(30, d, 22)
+    res = []
+    for bf16 in (0, 1):
+        spec = {"type": "ivf_flat", "dim": d, "metric": metric,
+                "nlist": nlist, "nprobe": 4, "seed": 3, "coarse_bf16": bf16}
+        eng = HipEngine(spec=spec)
+        eng.set_trained(cent)
+        eng.add(xb)
+        res.append(eng.search(q, 8))
+    np.testing.assert_array_equal(res[0][1], res[1][1])
+    np.testing.assert_array_equal(res[0][0], res[1][0])
+
+
+def test_coarse_bf16_trains():
+    # full train/add/search with bf16 k-means assignment: sane recall
+    d, nlist = 32, 8
+    cent, xb = _clustered(nlist, 400, d, seed=23)
+    spec = {"type": "ivfpq", "dim": d, "metric": L2, "nlist": nlist, "m": 8,
+            "nprobe": 8, "seed": 3, "coarse_bf16": 1}
+    eng = HipEngine(spec=spec)
+    eng.train(xb)
+    eng.add(xb)
+    q = xb[:20] + 0.01 * _rand(20, d, 24)
+    D, I = eng.search(q, 5)
+    hits = (I[:, 0] == np.arange(20)).mean()
+    assert hits > 0.9, f"bf16-trained recall {hits}"
+
+
 def test_chunked_assign_matches_unchunked():
     # ws_mb=1 forces the multi-chunk assign/coarse paths (the round-1
     # negative-OOB regression lived there): results must be identical to
