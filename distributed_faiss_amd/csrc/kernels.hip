@@ -761,6 +761,55 @@ __device__ void ivf_scan_body(
         }
       }
     }
+  } else if (FAM == 2) {
+    // SQ8: 8 lanes per row — a wave reads 8 consecutive rows as 1 KiB of
+    // CONTIGUOUS lines (the streaming-copy access pattern) instead of 64
+    // scattered line fragments. Reduction: fixed 3-step butterfly over the
+    // 8 lanes, mirrored op-for-op by the oracle (OracleIVFSQ._scan_one) —
+    // all lanes converge to the same bitwise sum.
+    {
+      const float *ubuf = fam, *vbuf = fam + d;
+      int g8 = threadIdx.x & 7, grp = threadIdx.x >> 3;  // 32 row-groups
+      for (long long base = s0; base < s1; base += 128) {
+        if (!REGSEL) sel_guard(s, k, 128);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          long long pos = base + (long long)u * 32 + grp;
+          bool valid = pos < s1;
+          float part = 0.f;
+          if (valid) {
+            const uint8_t *cp = codes + pos * (size_t)stride;
+            for (int t0 = g8 * 16; t0 < d; t0 += 128) {
+#pragma clang fp contract(off)
+              uint4 wv = *reinterpret_cast<const uint4 *>(cp + t0);
+              unsigned w0_ = wv.x, w1_ = wv.y, w2_ = wv.z, w3_ = wv.w;
+#pragma unroll
+              for (int b = 0; b < 16; ++b) {
+                if (t0 + b < d) {
+                  unsigned word = (b < 4) ? w0_ : (b < 8) ? w1_ : (b < 12) ? w2_ : w3_;
+                  float cf = DFANN_CVT_UB(word, b);
+                  int t = t0 + b;
+                  if (IS_IP) {
+                    part = part + (ubuf[t] + cf * vbuf[t]);
+                  } else {
+                    float diff = ubuf[t] - cf * vbuf[t];
+                    part = part + diff * diff;
+                  }
+                }
+              }
+            }
+          }
+          part += __shfl_xor(part, 4, 8);
+          part += __shfl_xor(part, 2, 8);
+          part += __shfl_xor(part, 1, 8);
+          float dist = IS_IP ? -(bias + part) : part;
+          if (valid && g8 == 0) {
+            if (REGSEL) loc.push(dist, (unsigned)pos);
+            else sel_try(s, dist, (unsigned)pos);
+          }
+        }
+      }
+    }
   } else if (REGSEL) {
     // independent rows per thread per iteration: their load batches
     // overlap, hiding HBM/L2 latency without any block synchronization.

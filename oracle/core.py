@@ -624,27 +624,42 @@ class OracleIVFSQ(_OracleIVFBase):
         if codes.shape[0] == 0:
             return None
         if self.qtype == "8bit":
-            # folded decode algebra, op-for-op the HIP scan kernel
-            # (csrc/kernels.hip DFANN_PROC16_SQ8): L2 diff = u - c*v with
-            # u = (q-cent-vmin) - 0.5*scale, v = scale; IP acc += u + c*v
-            # with u = q*vmin + 0.5*(q*scale), v = q*scale. Equal to the
-            # faiss codec in exact arithmetic; this is the shared fp32
-            # rounding order of the bit-exact tier.
+            # folded decode algebra + 8-lane butterfly order, op-for-op the
+            # HIP scan kernel (csrc/kernels.hip FAM==2 subgroup-8 branch):
+            # L2 diff = u - c*v with u = (q-cent-vmin) - 0.5*scale,
+            # v = scale; IP term = u + c*v with u = q*vmin + 0.5*(q*scale),
+            # v = q*scale. Lane l sums dims {l*16 + 128*c + b} sequentially;
+            # the 8 partials combine as ((p0+p4)+(p2+p6)) + ((p1+p5)+(p3+p7)).
+            # Equal to the faiss codec in exact arithmetic; this is the
+            # shared fp32 rounding order of the bit-exact tier.
             cf = codes.astype(np.float32)
             if self.metric == METRIC_L2:
                 r = (qi - self.centroids[li]).astype(np.float32)
                 u = (r - self.vmin) - np.float32(0.5) * self.scale
                 v = self.scale
-                acc = np.zeros(codes.shape[0], dtype=np.float32)
-                for t in range(self.d):
+
+                def term(t):
                     diff = u[t] - cf[:, t] * v[t]
-                    acc = acc + diff * diff
+                    return diff * diff
+            else:
+                qsc = (qi * self.scale).astype(np.float32)
+                uip = qi * self.vmin + np.float32(0.5) * qsc
+
+                def term(t):
+                    return uip[t] + cf[:, t] * qsc[t]
+
+            n = codes.shape[0]
+            parts = [np.zeros(n, dtype=np.float32) for _ in range(8)]
+            for lane in range(8):
+                for t0 in range(lane * 16, self.d, 128):
+                    for b in range(16):
+                        t = t0 + b
+                        if t < self.d:
+                            parts[lane] = parts[lane] + term(t)
+            acc = ((parts[0] + parts[4]) + (parts[2] + parts[6])) + (
+                (parts[1] + parts[5]) + (parts[3] + parts[7]))
+            if self.metric == METRIC_L2:
                 return acc, self.list_ids[li]
-            qsc = (qi * self.scale).astype(np.float32)
-            u = qi * self.vmin + np.float32(0.5) * qsc
-            acc = np.zeros(codes.shape[0], dtype=np.float32)
-            for t in range(self.d):
-                acc = acc + (u[t] + cf[:, t] * qsc[t])
             if bias is None:
                 bias = np.float32(seq_ip(qi, self.centroids[li][None, :])[0])
             return bias + acc, self.list_ids[li]
