@@ -191,6 +191,7 @@ struct dfann_index {
   int code_bytes = 0, stride = 0;
   int ws_mb = 512;  // chunk budget for key matrices (spec "ws_mb"; tests
                     // shrink it to force the multi-chunk paths)
+  int max_ppc = 256;  // k-means subsample cap per centroid (spec "max_ppc")
   bool coarse_bf16 = false;  // spec "coarse_bf16": assign/coarse GEMMs on
                              // bf16 MFMA (~16x f32 rate) — approximate
                              // ranking path for huge nlist
@@ -342,7 +343,7 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
                           int d, int metric, uint64_t seed, float *cent_out,
                           hipStream_t stream) {
   const int NITER = 25;
-  int64_t cap = (int64_t)kcent * 256;
+  int64_t cap = (int64_t)kcent * (h ? h->max_ppc : 256);
   DevBuf xt_buf;
   const float *xt = x;
   int64_t nt = n;
@@ -467,6 +468,7 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->ws_mb = (int)json_int(js, "ws_mb", 512);
   if (h->ws_mb < 1) h->ws_mb = 1;
   h->coarse_bf16 = json_int(js, "coarse_bf16", 0) != 0;
+  h->max_ppc = (int)json_int(js, "max_ppc", 256);
   if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
   if (h->type != T_FLAT && h->nlist <= 0) {
     delete h;
