@@ -770,19 +770,33 @@ __device__ void ivf_scan_body(
     {
       const float *ubuf = fam, *vbuf = fam + d;
       int g8 = threadIdx.x & 7, grp = threadIdx.x >> 3;  // 32 row-groups
+      const bool onechunk = d <= 128;  // lane covers one 16-B chunk
       for (long long base = s0; base < s1; base += 128) {
         if (!REGSEL) sel_guard(s, k, 128);
+        if (onechunk) {
+          // issue all 4 row-set loads up front: 4 independent HBM
+          // requests in flight per wave instead of 1 (latency hiding)
+          uint4 wv4[4];
+          bool val4[4];
 #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          long long pos = base + (long long)u * 32 + grp;
-          bool valid = pos < s1;
-          float part = 0.f;
-          if (valid) {
-            const uint8_t *cp = codes + pos * (size_t)stride;
-            for (int t0 = g8 * 16; t0 < d; t0 += 128) {
+          for (int u = 0; u < 4; ++u) {
+            long long pos = base + (long long)u * 32 + grp;
+            val4[u] = pos < s1;
+            int t0 = g8 * 16;
+            wv4[u] = (val4[u] && t0 < d)
+                         ? *reinterpret_cast<const uint4 *>(
+                               codes + pos * (size_t)stride + t0)
+                         : uint4{0, 0, 0, 0};
+          }
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            long long pos = base + (long long)u * 32 + grp;
+            float part = 0.f;
+            int t0 = g8 * 16;
+            if (t0 < d) {
 #pragma clang fp contract(off)
-              uint4 wv = *reinterpret_cast<const uint4 *>(cp + t0);
-              unsigned w0_ = wv.x, w1_ = wv.y, w2_ = wv.z, w3_ = wv.w;
+              unsigned w0_ = wv4[u].x, w1_ = wv4[u].y, w2_ = wv4[u].z,
+                       w3_ = wv4[u].w;
 #pragma unroll
               for (int b = 0; b < 16; ++b) {
                 if (t0 + b < d) {
@@ -798,14 +812,51 @@ __device__ void ivf_scan_body(
                 }
               }
             }
+            part += __shfl_xor(part, 4, 8);
+            part += __shfl_xor(part, 2, 8);
+            part += __shfl_xor(part, 1, 8);
+            float dist = IS_IP ? -(bias + part) : part;
+            if (val4[u] && g8 == 0) {
+              if (REGSEL) loc.push(dist, (unsigned)pos);
+              else sel_try(s, dist, (unsigned)pos);
+            }
           }
-          part += __shfl_xor(part, 4, 8);
-          part += __shfl_xor(part, 2, 8);
-          part += __shfl_xor(part, 1, 8);
-          float dist = IS_IP ? -(bias + part) : part;
-          if (valid && g8 == 0) {
-            if (REGSEL) loc.push(dist, (unsigned)pos);
-            else sel_try(s, dist, (unsigned)pos);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            long long pos = base + (long long)u * 32 + grp;
+            bool valid = pos < s1;
+            float part = 0.f;
+            if (valid) {
+              const uint8_t *cp = codes + pos * (size_t)stride;
+              for (int t0 = g8 * 16; t0 < d; t0 += 128) {
+#pragma clang fp contract(off)
+                uint4 wv = *reinterpret_cast<const uint4 *>(cp + t0);
+                unsigned w0_ = wv.x, w1_ = wv.y, w2_ = wv.z, w3_ = wv.w;
+#pragma unroll
+                for (int b = 0; b < 16; ++b) {
+                  if (t0 + b < d) {
+                    unsigned word = (b < 4) ? w0_ : (b < 8) ? w1_ : (b < 12) ? w2_ : w3_;
+                    float cf = DFANN_CVT_UB(word, b);
+                    int t = t0 + b;
+                    if (IS_IP) {
+                      part = part + (ubuf[t] + cf * vbuf[t]);
+                    } else {
+                      float diff = ubuf[t] - cf * vbuf[t];
+                      part = part + diff * diff;
+                    }
+                  }
+                }
+              }
+            }
+            part += __shfl_xor(part, 4, 8);
+            part += __shfl_xor(part, 2, 8);
+            part += __shfl_xor(part, 1, 8);
+            float dist = IS_IP ? -(bias + part) : part;
+            if (valid && g8 == 0) {
+              if (REGSEL) loc.push(dist, (unsigned)pos);
+              else sel_try(s, dist, (unsigned)pos);
+            }
           }
         }
       }
