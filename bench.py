@@ -70,6 +70,13 @@ WORKLOADS = {
         metric=1, nq=10_000, k=10, centers=50_000, sigma=0.5,
         sq_type="8bit",
     ),
+    # BASELINE configs[3]-SHAPED at single-GPU-buildable scale: d=768,
+    # m=64 (64 KB LUT in LDS), bf16-MFMA assign/coarse build path
+    "ivfpq_2m_d768_m64": dict(
+        type="ivfpq", d=768, n=2_000_000, nlist=2048, m=64, nbits=8,
+        metric=1, nq=10_000, k=10, centers=20_000, sigma=0.5,
+        coarse_bf16=1, max_ppc=64,
+    ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
         type="ivfpq", d=64, n=100_000, nlist=256, m=8, nbits=8,
@@ -171,7 +178,9 @@ def main():
 
     spec = {"type": cfg["type"], "dim": cfg["d"], "metric": metric,
             "nlist": cfg["nlist"], "m": cfg["m"], "nbits": cfg["nbits"],
-            "sq_type": cfg.get("sq_type", "fp16"), "nprobe": 1, "seed": 1234}
+            "sq_type": cfg.get("sq_type", "fp16"), "nprobe": 1, "seed": 1234,
+            "coarse_bf16": cfg.get("coarse_bf16", 0),
+            "max_ppc": cfg.get("max_ppc", 256)}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)

@@ -1149,6 +1149,15 @@ extern "C" __global__ __launch_bounds__(256) void k_pq_encode(
   }
 }
 
+// write subspace argmin results into the packed code column j
+extern "C" __global__ void k_codes_from_best(const int *__restrict__ best,
+                                             long long n, int j, int stride,
+                                             uint8_t *__restrict__ codes) {
+  long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; p < n; p += (long long)gridDim.x * blockDim.x)
+    codes[p * (size_t)stride + j] = (uint8_t)best[p];
+}
+
 // SQ encode (8bit / fp16) of residuals
 extern "C" __global__ void k_sq_encode(const float *__restrict__ resid,
                                        const float *__restrict__ vmin,

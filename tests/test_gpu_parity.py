@@ -169,9 +169,18 @@ def _bitexact_setup(typ, metric, sq_type=None, d=32, nlist=8, per=400):
     rng = np.random.default_rng(11)
     kwargs = {}
     if typ == "ivfpq":
+        # points = centroid + exact codebook words (+noise << codeword
+        # margins) so the per-subspace argmin is unambiguous: engine
+        # (GEMM-decomposition encode) and oracle (sequential encode) pick
+        # identical codes despite different rounding orders
         spec["m"] = 8
         cb = rng.standard_normal((8, 256, d // 8)).astype(np.float32)
         kwargs["codebooks"] = cb
+        lbl = rng.integers(0, nlist, xb.shape[0])
+        cw = rng.integers(0, 256, (xb.shape[0], 8))
+        dec = np.concatenate([cb[j][cw[:, j]] for j in range(8)], axis=1)
+        xb = (cent[lbl] + dec
+              + 1e-3 * rng.standard_normal(xb.shape)).astype(np.float32)
     if typ == "ivfsq":
         spec["sq_type"] = sq_type
         if sq_type == "8bit":
