@@ -77,6 +77,15 @@ WORKLOADS = {
         metric=1, nq=10_000, k=10, centers=20_000, sigma=0.5,
         coarse_bf16=1, max_ppc=64,
     ),
+    # THE HEADLINE: BASELINE.json configs[3] per-shard slice — 12.5M x 768
+    # IVFPQ m=64, nlist=65536/shard; at --gpus 8 the sharded DB is exactly
+    # the 100M x 768 configuration. bf16-MFMA assign/coarse for the build;
+    # k-means capped at 64 pts/centroid (~4.2M training rows/shard).
+    "ivfpq_100m8_d768_m64": dict(
+        type="ivfpq", d=768, n=12_500_000, nlist=65536, m=64, nbits=8,
+        metric=1, nq=10_000, k=10, centers=12_500, sigma=0.5,
+        coarse_bf16=1, max_ppc=64, ws_mb=2048,
+    ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
         type="ivfpq", d=64, n=100_000, nlist=256, m=8, nbits=8,
@@ -180,7 +189,8 @@ def main():
             "nlist": cfg["nlist"], "m": cfg["m"], "nbits": cfg["nbits"],
             "sq_type": cfg.get("sq_type", "fp16"), "nprobe": 1, "seed": 1234,
             "coarse_bf16": cfg.get("coarse_bf16", 0),
-            "max_ppc": cfg.get("max_ppc", 256)}
+            "max_ppc": cfg.get("max_ppc", 256),
+            "ws_mb": cfg.get("ws_mb", 512)}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)
