@@ -378,6 +378,7 @@ def main():
                 "recall_at_10": recall,
                 "batch": cfg["nq"],
                 "metric_space": "l2" if metric == 1 else "dot",
+                "coarse_dtype": "bf16" if cfg.get("coarse_bf16") else "f32",
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
