@@ -84,7 +84,7 @@ WORKLOADS = {
     "ivfpq_100m8_d768_m64": dict(
         type="ivfpq", d=768, n=12_500_000, nlist=65536, m=64, nbits=8,
         metric=1, nq=10_000, k=10, centers=12_500, sigma=0.5,
-        coarse_bf16=1, max_ppc=64, ws_mb=2048,
+        coarse_bf16=1, max_ppc=64, ws_mb=2048, pq_precomputed=1,
     ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
@@ -190,7 +190,8 @@ def main():
             "sq_type": cfg.get("sq_type", "fp16"), "nprobe": 1, "seed": 1234,
             "coarse_bf16": cfg.get("coarse_bf16", 0),
             "max_ppc": cfg.get("max_ppc", 256),
-            "ws_mb": cfg.get("ws_mb", 512)}
+            "ws_mb": cfg.get("ws_mb", 512),
+            "pq_precomputed": cfg.get("pq_precomputed", 0)}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)

@@ -192,7 +192,8 @@ struct dfann_index {
   int ws_mb = 512;  // chunk budget for key matrices (spec "ws_mb"; tests
                     // shrink it to force the multi-chunk paths)
   int max_ppc = 256;  // k-means subsample cap per centroid (spec "max_ppc")
-  bool coarse_bf16 = false;  // spec "coarse_bf16": assign/coarse GEMMs on
+  bool coarse_bf16 = false;
+  bool pq_pre = false;  // spec "pq_precomputed": PQ-L2 term2/term3 tables  // spec "coarse_bf16": assign/coarse GEMMs on
                              // bf16 MFMA (~16x f32 rate) — approximate
                              // ranking path for huge nlist
 
@@ -209,6 +210,7 @@ struct dfann_index {
   // workspace
   DevBuf ws1, ws2, ws3, ws4, ws5, ws_bf16a, ws_bf16b;
   DevBuf cent_bf16;
+  DevBuf term2, term3_ws, qn_ws;  // PQ-L2 precomputed tables
 
   // timing
   bool timing = false;
@@ -266,13 +268,11 @@ static void gemm_keys_b(dfann_index *h, const float *A, int64_t Mrows,
                        A, Mrows * K, h->ws_bf16a.as<unsigned short>());
     hipLaunchKernelGGL(k_gemm_bf16_nt, g, dim3(256), 0, stream,
                        h->ws_bf16a.as<unsigned short>(), B_bf16, keys,
-                       (int)Mrows, (int)N, K, K, K, (int)N);
+                       (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);
   } else {
     hipLaunchKernelGGL(k_gemm_nt, g, dim3(256), 0, stream, A, B, keys,
-                       (int)Mrows, (int)N, K, K, K, (int)N);
+                       (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);
   }
-  hipLaunchKernelGGL(k_score_key, grid1d(Mrows * N), dim3(256), 0, stream, keys,
-                     Mrows, N, N, qn, bn, mode);
   if (h && h->timing) {
     h->ev_end(e, stream, h->ev_gemm);
     h->gemm_flops += 2LL * Mrows * N * K;
@@ -287,14 +287,24 @@ static void gemm_keys(dfann_index *h, const float *A, int64_t Mrows,
   gemm_keys_b(h, A, Mrows, B, nullptr, N, K, bn, qn, mode, keys, stream);
 }
 
-// refresh the bf16 centroid image after (re)training / loading
+// refresh derived images after (re)training / loading: bf16 centroids,
+// PQ-L2 precomputed term2
 static void refresh_cent_bf16(dfann_index *h, hipStream_t stream) {
-  if (!h->coarse_bf16 || !h->centroids.p) return;
-  h->cent_bf16.ensure((size_t)h->nlist * h->d * 2);
-  hipLaunchKernelGGL(k_f32_to_bf16, grid1d((int64_t)h->nlist * h->d), dim3(256),
-                     0, stream, h->centroids.as<float>(),
-                     (long long)h->nlist * h->d,
-                     h->cent_bf16.as<unsigned short>());
+  if (h->coarse_bf16 && h->centroids.p) {
+    h->cent_bf16.ensure((size_t)h->nlist * h->d * 2);
+    hipLaunchKernelGGL(k_f32_to_bf16, grid1d((int64_t)h->nlist * h->d),
+                       dim3(256), 0, stream, h->centroids.as<float>(),
+                       (long long)h->nlist * h->d,
+                       h->cent_bf16.as<unsigned short>());
+  }
+  if (h->pq_pre && h->type == T_IVFPQ && h->metric == M_L2 &&
+      h->centroids.p && h->codebooks.p) {
+    h->term2.ensure((size_t)h->nlist * h->m * 256 * 4);
+    hipLaunchKernelGGL(k_pq_term2, grid1d((int64_t)h->nlist * h->m * 256),
+                       dim3(256), 0, stream, h->centroids.as<float>(),
+                       h->codebooks.as<float>(), h->nlist, h->m, h->dsub,
+                       h->term2.as<float>());
+  }
   HIP_CHECK(hipGetLastError());
 }
 
@@ -325,9 +335,9 @@ static void assign_rows(dfann_index *h, const float *x, int64_t n,
                 h->cent_bf16.as<unsigned short>(), nlist, h->d,
                 h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, keys,
                 stream);
-    hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream, keys, c,
-                       (long long)nlist, (long long)nlist, 0, bestv + s,
-                       assign_dev + s);
+    hipLaunchKernelGGL(k_assign_rowblock, dim3((unsigned)c), dim3(256), 0,
+                       stream, keys, c, (long long)nlist, (long long)nlist, 0,
+                       bestv + s, assign_dev + s);
   }
   HIP_CHECK(hipGetLastError());
 }
@@ -398,8 +408,8 @@ static void kmeans_device(dfann_index *h, const float *x, int64_t n, int kcent,
                   use_b16 ? cent_b16.as<unsigned short>() : nullptr, kcent, d,
                   cn.as<float>(), nullptr, metric == M_IP ? 0 : 1,
                   keys.as<float>(), stream);
-      hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream,
-                         keys.as<float>(), c, (long long)kcent,
+      hipLaunchKernelGGL(k_assign_rowblock, dim3((unsigned)c), dim3(256), 0,
+                         stream, keys.as<float>(), c, (long long)kcent,
                          (long long)kcent, 0, bestv.as<float>() + s,
                          asg.as<int>() + s);
     }
@@ -469,6 +479,7 @@ static dfann_index *create_from_spec(const std::string &js) {
   if (h->ws_mb < 1) h->ws_mb = 1;
   h->coarse_bf16 = json_int(js, "coarse_bf16", 0) != 0;
   h->max_ppc = (int)json_int(js, "max_ppc", 256);
+  h->pq_pre = json_int(js, "pq_precomputed", 0) != 0;
   if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
   if (h->type != T_FLAT && h->nlist <= 0) {
     delete h;
@@ -506,7 +517,7 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
   trace_point("train:coarse-kmeans", stream);
   rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(),
            stream);
-  refresh_cent_bf16(h, stream);
+  refresh_cent_bf16(h, stream);  // bf16 image for the assign below
   if (h->type == T_IVFPQ || (h->type == T_IVFSQ && h->sq8)) {
     DevBuf asg, resid;
     asg.ensure((size_t)n * 4);
@@ -549,6 +560,7 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
                          h->sq_scale.as<float>());
     }
   }
+  refresh_cent_bf16(h, stream);  // + term2 now that codebooks exist
   HIP_CHECK(hipStreamSynchronize(stream));
   h->trained = true;
 }
@@ -608,8 +620,8 @@ static void add_impl(dfann_index *h, int64_t n, const float *x,
           gemm_keys(nullptr, sub.as<float>() + s0 * h->dsub, c, cbj, 256,
                     h->dsub, cbn.as<float>(), nullptr, 1, h->ws1.as<float>(),
                     stream);
-          hipLaunchKernelGGL(k_assign_chunk, grid1d(c), dim3(256), 0, stream,
-                             h->ws1.as<float>(), c, (long long)256,
+          hipLaunchKernelGGL(k_assign_rowblock, dim3((unsigned)c), dim3(256),
+                             0, stream, h->ws1.as<float>(), c, (long long)256,
                              (long long)256, 0, bestv.as<float>() + s0,
                              best.as<int>() + s0);
         }
@@ -752,18 +764,40 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
     default:
       throw std::runtime_error("scan on flat index");
   }
+  bool use_pre = h->type == T_IVFPQ && h->metric == M_L2 && h->pq_pre &&
+                 h->term2.p;
+  if (use_pre) fam_floats = h->m * 256;  // LUT only, no rbuf
   size_t lds = (size_t)fam_floats * 4 + (rk ? REGSEL_LDS_BYTES : SEL_LDS_BYTES);
   if (lds > 160 * 1024)
     throw std::runtime_error("scan LDS over budget (m too large)");
   TimingEv e;
   if (h->timing) e = h->ev_begin(stream);
-  hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe)), dim3(256), lds,
-                     stream, q, h->centroids.as<float>(),
-                     h->codebooks.as<float>(), h->sq_vmin.as<float>(),
-                     h->sq_scale.as<float>(), probes, keys,
-                     h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
-                     (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride, cand_d,
-                     cand_p, fam_floats);
+  if (use_pre) {
+    // per-batch tables: q norms + term3
+    h->qn_ws.ensure((size_t)nq * 4);
+    h->term3_ws.ensure((size_t)nq * h->m * 256 * 4);
+    rownorms(q, nq, h->d, h->qn_ws.as<float>(), stream);
+    hipLaunchKernelGGL(k_pq_term3, grid1d(nq * (int64_t)h->m * 256), dim3(256),
+                       0, stream, q, h->codebooks.as<float>(), nq, h->m,
+                       h->dsub, h->term3_ws.as<float>());
+    auto pk = rk ? k_scan_pq_l2_pre_rk : k_scan_pq_l2_pre;
+    hipLaunchKernelGGL(pk, dim3((unsigned)(nq * nprobe)), dim3(256), lds,
+                       stream, q, h->centroids.as<float>(),
+                       h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                       h->sq_scale.as<float>(), probes, keys,
+                       h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
+                       (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
+                       cand_d, cand_p, fam_floats, h->term2.as<float>(),
+                       h->term3_ws.as<float>(), h->qn_ws.as<float>());
+  } else {
+    hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe)), dim3(256), lds,
+                       stream, q, h->centroids.as<float>(),
+                       h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                       h->sq_scale.as<float>(), probes, keys,
+                       h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
+                       (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
+                       cand_d, cand_p, fam_floats);
+  }
   if (h->timing) {
     h->ev_end(e, stream, h->ev_scan);
     // algorithmic units: sum of probed list lengths
@@ -1038,7 +1072,6 @@ extern "C" int dfann_set_trained(dfann_index *h, const float *centroids_host,
   HIP_CHECK(hipMemcpy(h->centroids.p, centroids_host,
                       (size_t)h->nlist * h->d * 4, hipMemcpyHostToDevice));
   rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
-  refresh_cent_bf16(h, 0);
   if (h->type == T_IVFPQ) {
     if (!codebooks_host) throw std::runtime_error("codebooks required");
     h->codebooks.ensure((size_t)h->m * 256 * h->dsub * 4);
@@ -1062,6 +1095,7 @@ extern "C" int dfann_set_trained(dfann_index *h, const float *centroids_host,
     HIP_CHECK(hipMemcpy(h->sq_scale.p, sc.data(), (size_t)h->d * 4,
                         hipMemcpyHostToDevice));
   }
+  refresh_cent_bf16(h, 0);
   HIP_CHECK(hipDeviceSynchronize());
   h->trained = true;
   API_END
@@ -1181,9 +1215,9 @@ extern "C" int dfann_load(const char *path, dfann_index **out) {
       load_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
       h->cnorm.ensure((size_t)h->nlist * 4);
       rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
-      refresh_cent_bf16(h, 0);
       if (h->type == T_IVFPQ)
         load_dev(f, h->codebooks, (size_t)h->m * 256 * h->dsub * 4);
+      refresh_cent_bf16(h, 0);
       if (h->type == T_IVFSQ && h->sq8) {
         load_dev(f, h->sq_vmin, (size_t)h->d * 4);
         load_dev(f, h->sq_vdiff, (size_t)h->d * 4);

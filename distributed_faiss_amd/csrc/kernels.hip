@@ -240,9 +240,22 @@ __device__ void regtopk_block_extract(RegTopK<K> &loc, int k, char *lds,
 #define GT 128
 #define GK 32
 
+__device__ __forceinline__ float gemm_key(float v, int row, int col,
+                                          const float *qn, const float *bn,
+                                          int mode) {
+  if (mode == 0) return -v;
+  if (mode == 1) return bn[col] - 2.0f * v;
+  if (mode == 2) return (qn[row] - 2.0f * v) + bn[col];
+  return v;
+}
+
+// mode: -1 raw ip; 0 key=-ip; 1 key=bn[col]-2ip; 2 key=(qn[row]-2ip)+bn[col]
+// (fused epilogue — formulas identical to k_score_key, which remains for
+// standalone use)
 extern "C" __global__ __launch_bounds__(256) void k_gemm_nt(
     const float *__restrict__ A, const float *__restrict__ B,
-    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc) {
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
+    const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
   __shared__ float sA[GT][GK + 1];
   __shared__ float sB[GT][GK + 1];
   int bi = blockIdx.y * GT;
@@ -280,14 +293,18 @@ extern "C" __global__ __launch_bounds__(256) void k_gemm_nt(
     int rit = (rg & 3) + 8 * (rg >> 2) + row_in_tile_base;
     {
       int row = bi + wr + rit, col = bj + wc + col_in_tile;
-      if (row < M && col < N) C[(size_t)row * ldc + col] = acc00[rg];
+      if (row < M && col < N)
+        C[(size_t)row * ldc + col] = gemm_key(acc00[rg], row, col, qn, bn, mode);
       col = bj + wc + 32 + col_in_tile;
-      if (row < M && col < N) C[(size_t)row * ldc + col] = acc01[rg];
+      if (row < M && col < N)
+        C[(size_t)row * ldc + col] = gemm_key(acc01[rg], row, col, qn, bn, mode);
       row = bi + wr + 32 + rit;
       col = bj + wc + col_in_tile;
-      if (row < M && col < N) C[(size_t)row * ldc + col] = acc10[rg];
+      if (row < M && col < N)
+        C[(size_t)row * ldc + col] = gemm_key(acc10[rg], row, col, qn, bn, mode);
       col = bj + wc + 32 + col_in_tile;
-      if (row < M && col < N) C[(size_t)row * ldc + col] = acc11[rg];
+      if (row < M && col < N)
+        C[(size_t)row * ldc + col] = gemm_key(acc11[rg], row, col, qn, bn, mode);
     }
   }
 }
@@ -324,7 +341,8 @@ extern "C" __global__ void k_f32_to_bf16(const float *__restrict__ in,
 
 extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
     const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
-    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc) {
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
+    const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
   __shared__ unsigned short sA[GB_T][GB_K + 8];
   __shared__ unsigned short sB[GB_T][GB_K + 8];
   int bi = blockIdx.y * GB_T;
@@ -368,7 +386,9 @@ extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
       for (int rg = 0; rg < 4; ++rg) {
         int row = bi + wr + ti * 16 + rrow + rg;
         int col = bj + wc + tj * 16 + li;
-        if (row < M && col < N) C[(size_t)row * ldc + col] = acc[ti][tj][rg];
+        if (row < M && col < N)
+          C[(size_t)row * ldc + col] =
+              gemm_key(acc[ti][tj][rg], row, col, qn, bn, mode);
       }
     }
   }
@@ -483,6 +503,45 @@ extern "C" __global__ void k_assign_init(float *best_v, int *best_i, long long n
   if (i < n) {
     best_v[i] = DFANN_FLT_MAX;
     best_i[i] = -1;
+  }
+}
+
+// block-per-row running argmin: 256 threads reduce a row of the key
+// matrix (the serial per-thread variant k_assign_chunk was the train-time
+// bottleneck at nlist=65536: 1 thread x 65536 sequential reads).
+// Ties: lowest column; running compare strict < keeps the earlier
+// (lower-col) chunk on ties.
+extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
+    const float *__restrict__ keys, long long rows, long long cols,
+    long long ldk, int col_base, float *__restrict__ best_v,
+    int *__restrict__ best_i) {
+  long long r = blockIdx.x;
+  if (r >= rows) return;
+  const float *kp = keys + r * ldk;
+  float bv = DFANN_FLT_MAX;
+  int bi = 0x7FFFFFFF;
+  for (long long c = threadIdx.x; c < cols; c += blockDim.x) {
+    float v = kp[c];
+    if (v < bv || (v == bv && (int)c < bi)) { bv = v; bi = (int)c; }
+  }
+#pragma unroll
+  for (int o = 32; o > 0; o >>= 1) {
+    float ov = __shfl_down(bv, o, 64);
+    int oi = __shfl_down(bi, o, 64);
+    if (ov < bv || (ov == bv && oi < bi)) { bv = ov; bi = oi; }
+  }
+  __shared__ float wv[4];
+  __shared__ int wi[4];
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  if (lane == 0) { wv[w] = bv; wi[w] = bi; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int i = 1; i < 4; ++i)
+      if (wv[i] < bv || (wv[i] == bv && wi[i] < bi)) { bv = wv[i]; bi = wi[i]; }
+    if (bv < best_v[r]) {
+      best_v[r] = bv;
+      best_i[r] = col_base + bi;
+    }
   }
 }
 
@@ -645,7 +704,7 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
   return acc;
 }
 
-template <int FAM, bool IS_IP, bool REGSEL>
+template <int FAM, bool IS_IP, bool REGSEL, bool PRE = false>
 __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
@@ -653,7 +712,10 @@ __device__ void ivf_scan_body(
     const float *__restrict__ keys, const uint8_t *__restrict__ codes,
     const int64_t *__restrict__ off, int nq, int nprobe, int d, int m,
     int dsub, int k, int stride, float *__restrict__ cand_d,
-    unsigned *__restrict__ cand_p, int fam_floats) {
+    unsigned *__restrict__ cand_p, int fam_floats,
+    const float *__restrict__ term2 = nullptr,
+    const float *__restrict__ term3 = nullptr,
+    const float *__restrict__ qn = nullptr) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float *fam = reinterpret_cast<float *>(smem);
   char *selbase = smem + (size_t)fam_floats * 4;
@@ -672,9 +734,19 @@ __device__ void ivf_scan_body(
   const float *qp = q + (long long)bq * d;
   float bias = 0.f;
   if (IS_IP) bias = -keys[(long long)bq * nprobe + bp];  // q . centroid
+  if (PRE)   // full coarse L2 distance: ||q||^2 + (|c|^2 - 2 q.c)
+    bias = qn[bq] + keys[(long long)bq * nprobe + bp];
 
   // --- stage FAM region ---
-  if (FAM == 0) {
+  if (FAM == 0 && PRE) {
+    // LUT from the precomputed tables: 2 row reads instead of the whole
+    // codebook; dist = bias + sum_j LUT
+    float *lut = fam;
+    const float *t2 = term2 + (size_t)L * m * 256;
+    const float *t3 = term3 + (size_t)bq * m * 256;
+    for (int e = threadIdx.x; e < m * 256; e += blockDim.x)
+      lut[e] = t2[e] - 2.0f * t3[e];
+  } else if (FAM == 0) {
     // rbuf (d floats) AFTER the LUT
     float *lut = fam;
     float *rbuf = fam + (size_t)m * 256;
@@ -876,7 +948,7 @@ __device__ void ivf_scan_body(
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
           float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
-          float dist = IS_IP ? -(bias + acc) : acc;
+          float dist = IS_IP ? -(bias + acc) : (PRE ? bias + acc : acc);
           loc.push(dist, (unsigned)pos);
         }
       }
@@ -890,7 +962,7 @@ __device__ void ivf_scan_body(
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
           float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
-          float dist = IS_IP ? -(bias + acc) : acc;
+          float dist = IS_IP ? -(bias + acc) : (PRE ? bias + acc : acc);
           sel_try(s, dist, (unsigned)pos);
         }
       }
@@ -923,6 +995,24 @@ __device__ void ivf_scan_body(
                                       dsub, k, stride, cand_d, cand_p,         \
                                       fam_floats);                             \
   }
+
+#define INSTANTIATE_SCAN_PRE(NAME, REGSEL)                                     \
+  extern "C" __global__ __launch_bounds__(256) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
+      unsigned *cand_p, int fam_floats, const float *term2,                    \
+      const float *term3, const float *qn) {                                   \
+    ivf_scan_body<0, false, REGSEL, true>(q, cent, cb, sq_vmin, sq_scale,      \
+                                          probes, keys, codes, off, nq,        \
+                                          nprobe, d, m, dsub, k, stride,       \
+                                          cand_d, cand_p, fam_floats, term2,   \
+                                          term3, qn);                          \
+  }
+
+INSTANTIATE_SCAN_PRE(k_scan_pq_l2_pre, false)
+INSTANTIATE_SCAN_PRE(k_scan_pq_l2_pre_rk, true)
 
 INSTANTIATE_SCAN(k_scan_pq_l2, 0, false, false)
 INSTANTIATE_SCAN(k_scan_pq_ip, 0, true, false)
@@ -1180,6 +1270,52 @@ extern "C" __global__ void k_sq_encode(const float *__restrict__ resid,
       c = c < 0 ? 0 : (c > 255 ? 255 : c);
       codes[r * (size_t)stride + t] = (uint8_t)c;
     }
+  }
+}
+
+// PQ-L2 precomputed tables (faiss IndexIVFPQ use_precomputed_table
+// restated): dist(q, code | L) = [qn + coarse_key(q,L)] + sum_j LUT[j][k]
+// with LUT[j][k] = term2[L][j][k] - 2*term3[q][j][k],
+// term2 = ||cb_jk||^2 + 2 c_{L,j}.cb_jk   (per index, nlist*m*256 fp32),
+// term3 = q_j . cb_jk                     (per query batch).
+// Cuts the scan block's LUT-build traffic from the full m*256*dsub
+// codebook (786 KB at m=64,d=768) to 2 table rows (128 KB).
+extern "C" __global__ void k_pq_term2(const float *__restrict__ cent,
+                                      const float *__restrict__ cb, int nlist,
+                                      int m, int dsub,
+                                      float *__restrict__ term2) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = (long long)nlist * m * 256;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    int k = (int)(i & 255);
+    int j = (int)((i >> 8) % m);
+    long long L = i / (256LL * m);
+    const float *cbe = cb + ((size_t)j * 256 + k) * dsub;
+    const float *cj = cent + L * (size_t)m * dsub + (size_t)j * dsub;
+    float nrm = 0.f, dot = 0.f;
+    for (int t = 0; t < dsub; ++t) {
+      nrm += cbe[t] * cbe[t];
+      dot += cj[t] * cbe[t];
+    }
+    term2[i] = nrm + 2.0f * dot;
+  }
+}
+
+extern "C" __global__ void k_pq_term3(const float *__restrict__ q,
+                                      const float *__restrict__ cb,
+                                      long long nq, int m, int dsub,
+                                      float *__restrict__ term3) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long total = nq * m * 256;
+  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
+    int k = (int)(i & 255);
+    int j = (int)((i >> 8) % m);
+    long long qi = i / (256LL * m);
+    const float *cbe = cb + ((size_t)j * 256 + k) * dsub;
+    const float *qs = q + qi * (size_t)m * dsub + (size_t)j * dsub;
+    float dot = 0.f;
+    for (int t = 0; t < dsub; ++t) dot += qs[t] * cbe[t];
+    term3[i] = dot;
   }
 }
 

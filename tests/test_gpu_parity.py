@@ -277,6 +277,57 @@ def test_coarse_bf16_trains():
     assert hits > 0.9, f"bf16-trained recall {hits}"
 
 
+@pytest.mark.parametrize("metric", [IP, L2])
+def test_large_k_and_nprobe_bitonic_paths(metric):
+    # k=64 / nprobe=32 exercise the LDS-bitonic selection kernels (the
+    # register path covers only k<=16)
+    d, n = 32, 5000
+    xb, q = _rand(n, d, 50), _rand(15, d, 51)
+    spec = {"type": "ivf_flat", "dim": d, "metric": metric, "nlist": 32,
+            "nprobe": 32, "seed": 9}
+    eng = HipEngine(spec=spec)
+    eng.train(xb)
+    eng.add(xb)
+    D, I = eng.search(q, 64)
+    orc = _mk_oracle_with_engine_artifacts(eng, spec, xb)
+    Do, Io = orc.search(q, 64)
+    m = (I == Io).mean()
+    assert m > 0.98, f"id match {m}"
+    same = I == Io
+    np.testing.assert_allclose(D[same], Do[same], rtol=1e-4, atol=1e-4)
+    # flat large-k too
+    fe = HipEngine(spec={"type": "flat", "dim": d, "metric": metric})
+    fe.train(xb)
+    fe.add(xb)
+    Df, If = fe.search(q, 100)
+    fo = make_oracle_engine({"type": "flat", "dim": d, "metric": metric})
+    fo.add(xb)
+    Dfo, Ifo = fo.search(q, 100)
+    assert (If == Ifo).mean() > 0.99
+    np.testing.assert_allclose(Df, Dfo, rtol=1e-4, atol=1e-4)
+
+
+def test_pq_precomputed_table_path():
+    # faiss-style term2/term3 PQ-L2 scan vs the direct-LUT path: same
+    # math, different rounding — ids must agree, distances within 1e-4
+    d, nlist, m = 64, 16, 8
+    cent, xb = _clustered(nlist, 400, d, seed=31)
+    q = xb[::13][:25] + 0.01 * _rand(25, d, 32)
+    res = []
+    for pre in (0, 1):
+        spec = {"type": "ivfpq", "dim": d, "metric": L2, "nlist": nlist,
+                "m": m, "nprobe": 8, "seed": 5, "pq_precomputed": pre}
+        eng = HipEngine(spec=spec)
+        eng.train(xb)
+        eng.add(xb)
+        res.append(eng.search(q, 10))
+    (D0, I0), (D1, I1) = res
+    agree = (I0 == I1).mean()
+    assert agree > 0.99, f"pre vs direct id agreement {agree}"
+    same = I0 == I1
+    np.testing.assert_allclose(D0[same], D1[same], rtol=1e-4, atol=1e-4)
+
+
 def test_chunked_assign_matches_unchunked():
     # ws_mb=1 forces the multi-chunk assign/coarse paths (the round-1
     # negative-OOB regression lived there): results must be identical to
