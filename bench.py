@@ -84,7 +84,7 @@ WORKLOADS = {
     "ivfpq_100m8_d768_m64": dict(
         type="ivfpq", d=768, n=12_500_000, nlist=65536, m=64, nbits=8,
         metric=1, nq=10_000, k=10, centers=12_500, sigma=0.5,
-        coarse_bf16=1, max_ppc=64, ws_mb=2048, pq_precomputed=1,
+        coarse_bf16=1, max_ppc=64, ws_mb=2048,
     ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(

@@ -309,16 +309,22 @@ def test_large_k_and_nprobe_bitonic_paths(metric):
 
 def test_pq_precomputed_table_path():
     # faiss-style term2/term3 PQ-L2 scan vs the direct-LUT path: same
-    # math, different rounding — ids must agree, distances within 1e-4
+    # math, different rounding — ids must agree, distances within 1e-4.
+    # Artifacts are SHARED (train once) so only the scan formula differs
+    # (separate trainings differ in fp32-atomic rounding).
     d, nlist, m = 64, 16, 8
     cent, xb = _clustered(nlist, 400, d, seed=31)
     q = xb[::13][:25] + 0.01 * _rand(25, d, 32)
+    trainer = HipEngine(spec={"type": "ivfpq", "dim": d, "metric": L2,
+                              "nlist": nlist, "m": m, "nprobe": 8, "seed": 5})
+    trainer.train(xb)
+    cents, cbs = trainer.get_centroids(), trainer.get_codebooks()
     res = []
     for pre in (0, 1):
         spec = {"type": "ivfpq", "dim": d, "metric": L2, "nlist": nlist,
                 "m": m, "nprobe": 8, "seed": 5, "pq_precomputed": pre}
         eng = HipEngine(spec=spec)
-        eng.train(xb)
+        eng.set_trained(cents, cbs)
         eng.add(xb)
         res.append(eng.search(q, 10))
     (D0, I0), (D1, I1) = res
