@@ -337,42 +337,63 @@ extern "C" __global__ void k_f32_to_bf16(const float *__restrict__ in,
 }
 
 #define GB_T 128
-#define GB_K 32
+#define GB_K 64
 
 extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
     const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
     float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
     const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
+  // +8 pad keeps each row's 8-element groups 16-B aligned and breaks the
+  // power-of-2 bank stride for the fragment reads
   __shared__ unsigned short sA[GB_T][GB_K + 8];
   __shared__ unsigned short sB[GB_T][GB_K + 8];
   int bi = blockIdx.y * GB_T;
   int bj = blockIdx.x * GB_T;
   int tid = threadIdx.x;
   int lane = tid & 63, w = tid >> 6;
-  // wave w covers rows [wr, wr+64) x cols [wc, wc+64): 4x4 tiles of 16x16
   int wr = (w >> 1) * 64, wc = (w & 1) * 64;
   f32x4 acc[4][4] = {};
-  int li = lane & 15;           // fragment row/col within tile
-  int ke = (lane >> 4) * 8;     // k-offset of this lane's 8 elements
+  int li = lane & 15;
+  int ke = (lane >> 4) * 8;
+  // vectorized staging: 8 bf16 (16 B) per load; GB_T*GB_K/8 = 1024 groups
   for (int k0 = 0; k0 < K; k0 += GB_K) {
-    for (int e = tid; e < GB_T * GB_K; e += 256) {
-      int r = e >> 5, c = e & 31;
-      sA[r][c] = (bi + r < M && k0 + c < K)
-                     ? A[(size_t)(bi + r) * lda + k0 + c] : (unsigned short)0;
-      sB[r][c] = (bj + r < N && k0 + c < K)
-                     ? B[(size_t)(bj + r) * ldb + k0 + c] : (unsigned short)0;
+    for (int e8 = tid; e8 < GB_T * (GB_K / 8); e8 += 256) {
+      int r = e8 >> 3, c8 = (e8 & 7) * 8;  // GB_K/8 == 8
+      uint4 va = {0, 0, 0, 0}, vb = {0, 0, 0, 0};
+      if (bi + r < M) {
+        if (k0 + c8 + 7 < K) {
+          va = *reinterpret_cast<const uint4 *>(&A[(size_t)(bi + r) * lda + k0 + c8]);
+        } else {
+          unsigned short tmp[8] = {};
+          for (int t = 0; t < 8; ++t)
+            if (k0 + c8 + t < K) tmp[t] = A[(size_t)(bi + r) * lda + k0 + c8 + t];
+          va = *reinterpret_cast<const uint4 *>(tmp);
+        }
+      }
+      if (bj + r < N) {
+        if (k0 + c8 + 7 < K) {
+          vb = *reinterpret_cast<const uint4 *>(&B[(size_t)(bj + r) * ldb + k0 + c8]);
+        } else {
+          unsigned short tmp[8] = {};
+          for (int t = 0; t < 8; ++t)
+            if (k0 + c8 + t < K) tmp[t] = B[(size_t)(bj + r) * ldb + k0 + c8 + t];
+          vb = *reinterpret_cast<const uint4 *>(tmp);
+        }
+      }
+      *reinterpret_cast<uint4 *>(&sA[r][c8]) = va;
+      *reinterpret_cast<uint4 *>(&sB[r][c8]) = vb;
     }
     __syncthreads();
-    // one mfma_f32_16x16x32_bf16 covers a full K=32 slice across the wave
-    // (lane l holds k = (l>>4)*8 + 0..7), so GB_K == 32 needs exactly one
-    // MFMA per (ti,tj)
 #pragma unroll
-    for (int ti = 0; ti < 4; ++ti) {
-      bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&sA[wr + ti * 16 + li][ke]);
+    for (int kk = 0; kk < GB_K; kk += 32) {
 #pragma unroll
-      for (int tj = 0; tj < 4; ++tj) {
-        bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(&sB[wc + tj * 16 + li][ke]);
-        acc[ti][tj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj], 0, 0, 0);
+      for (int ti = 0; ti < 4; ++ti) {
+        bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&sA[wr + ti * 16 + li][kk + ke]);
+#pragma unroll
+        for (int tj = 0; tj < 4; ++tj) {
+          bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(&sB[wc + tj * 16 + li][kk + ke]);
+          acc[ti][tj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
