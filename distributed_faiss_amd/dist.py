@@ -60,18 +60,17 @@ def merge_gathered(Dall, Iall, k, maximize):
     (D, shard_idx, local_ids): shard_idx/local_ids map each winner back to
     its shard and per-shard id, mirroring client.py:290,297-298."""
     if Dall.is_cuda:
+        import torch
+
         from .hip_engine import merge_topk_dev
 
         S, nq, kk = Dall.shape
         Dm, slots = merge_topk_dev(Dall, Iall, k, maximize)
-        slots = slots.cpu().numpy()
-        # slot = s*nq*k + q*k + j
-        s_idx = slots // (nq * kk)
-        j_idx = slots % kk
-        q_idx = (slots // kk) % nq
-        Ia = Iall.cpu().numpy()
-        local = Ia[s_idx, q_idx, j_idx]
-        return Dm.cpu().numpy(), s_idx, local
+        # slot = s*nq*k + q*k + j IS the flat index into Iall — decode on
+        # the GPU (host numpy here cost ~half the step time at 10k batch)
+        s_idx = torch.div(slots, nq * kk, rounding_mode="floor")
+        local = Iall.reshape(-1)[slots]
+        return Dm.cpu().numpy(), s_idx.cpu().numpy(), local.cpu().numpy()
     # CPU (gloo tests): numpy restatement
     Da = Dall.numpy()
     Ia = Iall.numpy()
