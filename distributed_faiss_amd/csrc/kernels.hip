@@ -339,14 +339,16 @@ extern "C" __global__ void k_f32_to_bf16(const float *__restrict__ in,
 #define GB_T 128
 #define GB_K 64
 
+// Double-buffered, T14-scheduled (cdna_hip_programming.md §5.5 T14 /
+// Guideline 15): tile t+1's global loads are ISSUED before tile t's MFMA
+// phase and their LDS write lands after, behind the other buffer — HBM
+// latency hides under the MFMAs, one barrier per K-tile.
 extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
     const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
     float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
     const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
-  // +8 pad keeps each row's 8-element groups 16-B aligned and breaks the
-  // power-of-2 bank stride for the fragment reads
-  __shared__ unsigned short sA[GB_T][GB_K + 8];
-  __shared__ unsigned short sB[GB_T][GB_K + 8];
+  __shared__ unsigned short sA[2][GB_T][GB_K + 8];
+  __shared__ unsigned short sB[2][GB_T][GB_K + 8];
   int bi = blockIdx.y * GB_T;
   int bj = blockIdx.x * GB_T;
   int tid = threadIdx.x;
@@ -355,49 +357,82 @@ extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
   f32x4 acc[4][4] = {};
   int li = lane & 15;
   int ke = (lane >> 4) * 8;
-  // vectorized staging: 8 bf16 (16 B) per load; GB_T*GB_K/8 = 1024 groups
-  for (int k0 = 0; k0 < K; k0 += GB_K) {
-    for (int e8 = tid; e8 < GB_T * (GB_K / 8); e8 += 256) {
-      int r = e8 >> 3, c8 = (e8 & 7) * 8;  // GB_K/8 == 8
-      uint4 va = {0, 0, 0, 0}, vb = {0, 0, 0, 0};
-      if (bi + r < M) {
-        if (k0 + c8 + 7 < K) {
-          va = *reinterpret_cast<const uint4 *>(&A[(size_t)(bi + r) * lda + k0 + c8]);
-        } else {
-          unsigned short tmp[8] = {};
-          for (int t = 0; t < 8; ++t)
-            if (k0 + c8 + t < K) tmp[t] = A[(size_t)(bi + r) * lda + k0 + c8 + t];
-          va = *reinterpret_cast<const uint4 *>(tmp);
-        }
-      }
-      if (bj + r < N) {
-        if (k0 + c8 + 7 < K) {
-          vb = *reinterpret_cast<const uint4 *>(&B[(size_t)(bj + r) * ldb + k0 + c8]);
-        } else {
-          unsigned short tmp[8] = {};
-          for (int t = 0; t < 8; ++t)
-            if (k0 + c8 + t < K) tmp[t] = B[(size_t)(bj + r) * ldb + k0 + c8 + t];
-          vb = *reinterpret_cast<const uint4 *>(tmp);
-        }
-      }
-      *reinterpret_cast<uint4 *>(&sA[r][c8]) = va;
-      *reinterpret_cast<uint4 *>(&sB[r][c8]) = vb;
-    }
-    __syncthreads();
+  // each thread stages 4 16-B groups per operand per tile
+  int r_[4], c8_[4];
 #pragma unroll
-    for (int kk = 0; kk < GB_K; kk += 32) {
-#pragma unroll
-      for (int ti = 0; ti < 4; ++ti) {
-        bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&sA[wr + ti * 16 + li][kk + ke]);
-#pragma unroll
-        for (int tj = 0; tj < 4; ++tj) {
-          bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(&sB[wc + tj * 16 + li][kk + ke]);
-          acc[ti][tj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj], 0, 0, 0);
-        }
-      }
-    }
-    __syncthreads();
+  for (int g = 0; g < 4; ++g) {
+    int e8 = tid + g * 256;
+    r_[g] = e8 >> 3;
+    c8_[g] = (e8 & 7) * 8;
   }
+  uint4 va[4], vb[4];
+
+#define GB_LOAD(K0)                                                            \
+  _Pragma("unroll") for (int g = 0; g < 4; ++g) {                              \
+    int r = r_[g], c8 = c8_[g];                                                \
+    uint4 x = {0, 0, 0, 0}, y = {0, 0, 0, 0};                                  \
+    if (bi + r < M) {                                                          \
+      if ((K0) + c8 + 7 < K) {                                                 \
+        x = *reinterpret_cast<const uint4 *>(                                  \
+            &A[(size_t)(bi + r) * lda + (K0) + c8]);                           \
+      } else {                                                                 \
+        unsigned short tmp[8] = {};                                            \
+        for (int t = 0; t < 8; ++t)                                            \
+          if ((K0) + c8 + t < K)                                               \
+            tmp[t] = A[(size_t)(bi + r) * lda + (K0) + c8 + t];                \
+        x = *reinterpret_cast<const uint4 *>(tmp);                             \
+      }                                                                        \
+    }                                                                          \
+    if (bj + r < N) {                                                          \
+      if ((K0) + c8 + 7 < K) {                                                 \
+        y = *reinterpret_cast<const uint4 *>(                                  \
+            &B[(size_t)(bj + r) * ldb + (K0) + c8]);                           \
+      } else {                                                                 \
+        unsigned short tmp[8] = {};                                            \
+        for (int t = 0; t < 8; ++t)                                            \
+          if ((K0) + c8 + t < K)                                               \
+            tmp[t] = B[(size_t)(bj + r) * ldb + (K0) + c8 + t];                \
+        y = *reinterpret_cast<const uint4 *>(tmp);                             \
+      }                                                                        \
+    }                                                                          \
+    va[g] = x;                                                                 \
+    vb[g] = y;                                                                 \
+  }
+
+#define GB_WRITE(BUF)                                                          \
+  _Pragma("unroll") for (int g = 0; g < 4; ++g) {                              \
+    *reinterpret_cast<uint4 *>(&sA[BUF][r_[g]][c8_[g]]) = va[g];               \
+    *reinterpret_cast<uint4 *>(&sB[BUF][r_[g]][c8_[g]]) = vb[g];               \
+  }
+
+#define GB_MFMA(BUF)                                                           \
+  _Pragma("unroll") for (int kk = 0; kk < GB_K; kk += 32) {                    \
+    _Pragma("unroll") for (int ti = 0; ti < 4; ++ti) {                         \
+      bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(                           \
+          &sA[BUF][wr + ti * 16 + li][kk + ke]);                               \
+      _Pragma("unroll") for (int tj = 0; tj < 4; ++tj) {                       \
+        bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(                         \
+            &sB[BUF][wc + tj * 16 + li][kk + ke]);                             \
+        acc[ti][tj] =                                                          \
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj],       \
+                                                    0, 0, 0);                  \
+      }                                                                        \
+    }                                                                          \
+  }
+
+  GB_LOAD(0)
+  GB_WRITE(0)
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = GB_K; k0 < K; k0 += GB_K) {
+    GB_LOAD(k0)      // in flight during the MFMAs below
+    GB_MFMA(cur)
+    GB_WRITE(cur ^ 1)  // waits the loads; other buffer, so no barrier first
+    __syncthreads();
+    cur ^= 1;
+  }
+  GB_MFMA(cur)
+
   int rrow = (lane >> 4) * 4;
 #pragma unroll
   for (int ti = 0; ti < 4; ++ti) {
@@ -413,6 +448,9 @@ extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
       }
     }
   }
+#undef GB_LOAD
+#undef GB_WRITE
+#undef GB_MFMA
 }
 
 // ---------------------------------------------------------------------------
