@@ -82,7 +82,15 @@ class IndexClient:
             self.sub_indexes: List[IndexServer] = list(servers)
         else:
             machine_ports = IndexClient.read_server_list(server_list_path)
-            self.sub_indexes = [resolve_inproc_server(port) for _host, port in machine_ports]
+            self.sub_indexes = []
+            for host, port in machine_ports:
+                try:
+                    self.sub_indexes.append(resolve_inproc_server(port))
+                except ConnectionError:
+                    # multi-node: fall back to the TCP transport (rpc.py)
+                    from .rpc import TcpClient
+
+                    self.sub_indexes.append(TcpClient(host, port))
         self.num_indexes = len(self.sub_indexes)
 
         index_ranks = [idx.get_rank() for idx in self.sub_indexes]
@@ -319,7 +327,11 @@ class IndexClient:
         self.pool.map(lambda idx: idx.set_omp_num_threads(num_threads), self.sub_indexes)
 
     def close(self):
-        pass  # no sockets to close in-process
+        # TCP proxies own sockets; in-process servers have no close()
+        for idx in self.sub_indexes:
+            close_fn = idx.__dict__.get("sock") and getattr(idx, "close", None)
+            if close_fn:
+                close_fn()
 
     def get_num_servers(self):
         return self.num_indexes
