@@ -597,7 +597,7 @@ static void add_impl(dfann_index *h, int64_t n, const float *x,
     if (h->type == T_IVFPQ) {
       // per-subspace encode as a distance GEMM + running argmin — the
       // GEMM path runs ~20x the naive per-thread argmin kernel
-      // (k_pq_encode is kept for reference in kernels.hip)
+      // runs ~20x the naive per-thread argmin kernel
       DevBuf sub, best, bestv, cbn;
       sub.ensure((size_t)n * h->dsub * 4);
       best.ensure((size_t)n * 4);

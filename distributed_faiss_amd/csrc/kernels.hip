@@ -250,8 +250,7 @@ __device__ __forceinline__ float gemm_key(float v, int row, int col,
 }
 
 // mode: -1 raw ip; 0 key=-ip; 1 key=bn[col]-2ip; 2 key=(qn[row]-2ip)+bn[col]
-// (fused epilogue — formulas identical to k_score_key, which remains for
-// standalone use)
+// (fused epilogue)
 extern "C" __global__ __launch_bounds__(256) void k_gemm_nt(
     const float *__restrict__ A, const float *__restrict__ B,
     float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
@@ -477,29 +476,6 @@ extern "C" __global__ __launch_bounds__(256) void k_rownorm(
 }
 
 // ---------------------------------------------------------------------------
-// score epilogue: minimize-keys from raw inner products
-// mode 0: key = -ip (IP);  1: key = bn[j] - 2*ip (L2 rank);
-// mode 2: key = (qn[i] - 2*ip) + bn[j] (full squared L2, oracle op order)
-// ---------------------------------------------------------------------------
-
-extern "C" __global__ void k_score_key(float *__restrict__ C, long long rows,
-                                       long long cols, long long ldc,
-                                       const float *__restrict__ qn,
-                                       const float *__restrict__ bn, int mode) {
-  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  long long total = rows * cols;
-  for (; i < total; i += (long long)gridDim.x * blockDim.x) {
-    long long r = i / cols, c = i % cols;
-    float v = C[r * ldc + c];
-    float key;
-    if (mode == 0) key = -v;
-    else if (mode == 1) key = bn[c] - 2.0f * v;
-    else key = (qn[r] - 2.0f * v) + bn[c];
-    C[r * ldc + c] = key;
-  }
-}
-
-// ---------------------------------------------------------------------------
 // top-k per row over a key matrix (stream-select), pos = base + col.
 // out_d/out_p: (rows, k). grid.x = rows. dynamic LDS: SEL_LDS_BYTES.
 // ---------------------------------------------------------------------------
@@ -566,8 +542,8 @@ extern "C" __global__ void k_assign_init(float *best_v, int *best_i, long long n
 }
 
 // block-per-row running argmin: 256 threads reduce a row of the key
-// matrix (the serial per-thread variant k_assign_chunk was the train-time
-// bottleneck at nlist=65536: 1 thread x 65536 sequential reads).
+// matrix (a serial per-thread variant was the train-time bottleneck at
+// nlist=65536: 1 thread x 65536 sequential reads).
 // Ties: lowest column; running compare strict < keeps the earlier
 // (lower-col) chunk on ties.
 extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
@@ -602,24 +578,6 @@ extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
       best_i[r] = col_base + bi;
     }
   }
-}
-
-extern "C" __global__ void k_assign_chunk(const float *__restrict__ keys,
-                                          long long rows, long long cols,
-                                          long long ldk, int col_base,
-                                          float *__restrict__ best_v,
-                                          int *__restrict__ best_i) {
-  long long r = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (r >= rows) return;
-  const float *kp = keys + r * ldk;
-  float bv = best_v[r];
-  int bi = best_i[r];
-  for (long long c = 0; c < cols; ++c) {
-    float v = kp[c];
-    if (v < bv) { bv = v; bi = col_base + (int)c; }
-  }
-  best_v[r] = bv;
-  best_i[r] = bi;
 }
 
 // ---------------------------------------------------------------------------
@@ -1262,43 +1220,8 @@ extern "C" __global__ void k_residual(const float *__restrict__ x,
   }
 }
 
-// PQ encode: one thread per point; codebook subspace staged in LDS when it
-// fits. resid: (n, d) residuals; codes out: (n, stride) u8.
-extern "C" __global__ __launch_bounds__(256) void k_pq_encode(
-    const float *__restrict__ resid, const float *__restrict__ cb, long long n,
-    int d, int m, int dsub, int stride, uint8_t *__restrict__ codes) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float *scb = reinterpret_cast<float *>(smem);  // 256*dsub when staged
-  bool stage = (size_t)256 * dsub * 4 <= 48 * 1024;
-  long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  for (int j = 0; j < m; ++j) {
-    const float *cbj = cb + (size_t)j * 256 * dsub;
-    if (stage) {
-      __syncthreads();
-      for (int e = threadIdx.x; e < 256 * dsub; e += blockDim.x) scb[e] = cbj[e];
-      __syncthreads();
-    }
-    const float *cbs = stage ? scb : cbj;
-    if (p < n) {
-      const float *rs = resid + p * d + j * dsub;
-      float best = DFANN_FLT_MAX;
-      int bc = 0;
-      for (int c = 0; c < 256; ++c) {
-        const float *ce = cbs + (size_t)c * dsub;
-        float acc = 0.f;
-        for (int t = 0; t < dsub; ++t) {
-#pragma clang fp contract(off)
-          float diff = rs[t] - ce[t];
-          acc = acc + diff * diff;
-        }
-        if (acc < best) { best = acc; bc = c; }
-      }
-      codes[p * (size_t)stride + j] = (uint8_t)bc;
-    }
-  }
-}
-
-// write subspace argmin results into the packed code column j
+// write subspace argmin results into the packed code column j (PQ
+// encode = per-subspace distance GEMM + k_assign_rowblock + this)
 extern "C" __global__ void k_codes_from_best(const int *__restrict__ best,
                                              long long n, int j, int stride,
                                              uint8_t *__restrict__ codes) {
