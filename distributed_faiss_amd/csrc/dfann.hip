@@ -595,9 +595,8 @@ static void add_impl(dfann_index *h, int64_t n, const float *x,
                        h->centroids.as<float>(), asg.as<int>(), n, h->d,
                        resid.as<float>());
     if (h->type == T_IVFPQ) {
-      // per-subspace encode as a distance GEMM + running argmin — the
-      // GEMM path runs ~20x the naive per-thread argmin kernel
-      // runs ~20x the naive per-thread argmin kernel
+      // per-subspace encode as a distance GEMM + running argmin (~20x a
+      // naive per-thread 256-way argmin kernel)
       DevBuf sub, best, bestv, cbn;
       sub.ensure((size_t)n * h->dsub * 4);
       best.ensure((size_t)n * 4);
