@@ -726,16 +726,14 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                            int nprobe, const int32_t *probes,
                            const float *keys, int k, float *D, int64_t *I,
                            hipStream_t stream) {
-  // candidate arrays
-  h->ws3.ensure((size_t)nq * nprobe * k * 4);
-  h->ws4.ensure((size_t)nq * nprobe * k * 4);
-  float *cand_d = h->ws3.as<float>();
-  unsigned *cand_p = h->ws4.as<unsigned>();
+  // candidate arrays (sized after `fan` is chosen below)
+  float *cand_d;
+  unsigned *cand_p;
   int fam_floats;
   void (*kern)(const float *, const float *, const float *, const float *,
                const float *, const int *, const float *, const uint8_t *,
                const int64_t *, int, int, int, int, int, int, int, float *,
-               unsigned *, int) = nullptr;
+               unsigned *, int, int) = nullptr;
   bool ip = h->metric == M_IP;
   bool rk = use_regsel(k);
   switch (h->type) {
@@ -766,9 +764,20 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   bool use_pre = h->type == T_IVFPQ && h->metric == M_L2 && h->pq_pre &&
                  h->term2.p;
   if (use_pre) fam_floats = h->m * 256;  // LUT only, no rbuf
+  // segment fan: fill the chip when nq*nprobe is small. PQ blocks pay a
+  // per-block LUT build, so only the cheap-staging families fan out.
+  int fan = 1;
+  if (h->type != T_IVFPQ) {
+    int64_t blocks = nq * (int64_t)nprobe;
+    while (fan < 8 && blocks * fan < 65536) fan *= 2;
+  }
   size_t lds = (size_t)fam_floats * 4 + (rk ? REGSEL_LDS_BYTES : SEL_LDS_BYTES);
   if (lds > 160 * 1024)
     throw std::runtime_error("scan LDS over budget (m too large)");
+  h->ws3.ensure((size_t)nq * nprobe * fan * k * 4);
+  h->ws4.ensure((size_t)nq * nprobe * fan * k * 4);
+  cand_d = h->ws3.as<float>();
+  cand_p = h->ws4.as<unsigned>();
   TimingEv e;
   if (h->timing) e = h->ev_begin(stream);
   if (use_pre) {
@@ -789,13 +798,13 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        cand_d, cand_p, fam_floats, h->term2.as<float>(),
                        h->term3_ws.as<float>(), h->qn_ws.as<float>());
   } else {
-    hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe)), dim3(256), lds,
-                       stream, q, h->centroids.as<float>(),
+    hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe * fan)), dim3(256),
+                       lds, stream, q, h->centroids.as<float>(),
                        h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                        h->sq_scale.as<float>(), probes, keys,
                        h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
                        (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
-                       cand_d, cand_p, fam_floats);
+                       cand_d, cand_p, fam_floats, fan);
   }
   if (h->timing) {
     h->ev_end(e, stream, h->ev_scan);
@@ -812,11 +821,13 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   if (use_regsel(k)) {
     hipLaunchKernelGGL(k_merge_cand_rk, dim3((unsigned)nq), dim3(256),
                        REGSEL_LDS_BYTES + 128, stream, cand_d, cand_p, nq,
-                       nprobe * k, k, h->cr_ids.as<int64_t>(), ip ? 1 : 0, D, I);
+                       nprobe * fan * k, k, h->cr_ids.as<int64_t>(), ip ? 1 : 0,
+                       D, I);
   } else {
     hipLaunchKernelGGL(k_merge_cand, dim3((unsigned)nq), dim3(256),
-                       SEL_LDS_BYTES, stream, cand_d, cand_p, nq, nprobe * k, k,
-                       h->cr_ids.as<int64_t>(), ip ? 1 : 0, D, I);
+                       SEL_LDS_BYTES, stream, cand_d, cand_p, nq,
+                       nprobe * fan * k, k, h->cr_ids.as<int64_t>(), ip ? 1 : 0,
+                       D, I);
   }
   if (h->timing) h->ev_end(em, stream, h->ev_merge);
   HIP_CHECK(hipGetLastError());

@@ -732,15 +732,24 @@ __device__ void ivf_scan_body(
     unsigned *__restrict__ cand_p, int fam_floats,
     const float *__restrict__ term2 = nullptr,
     const float *__restrict__ term3 = nullptr,
-    const float *__restrict__ qn = nullptr) {
+    const float *__restrict__ qn = nullptr, int fan = 1) {
+  // fan > 1: each (query, probe) list is split into `fan` segments, one
+  // block per segment — long-list tail imbalance at low block counts.
+  // EXACT results: every segment's top-k contains the segment's global
+  // winners; the candidate merge sees (nprobe*fan*k) entries per query.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float *fam = reinterpret_cast<float *>(smem);
   char *selbase = smem + (size_t)fam_floats * 4;
   long long blk = blockIdx.x;
-  int bq = (int)(blk / nprobe), bp = (int)(blk % nprobe);
+  int bq = (int)(blk / ((long long)nprobe * fan));
+  int rest = (int)(blk % ((long long)nprobe * fan));
+  int bp = rest / fan, seg = rest % fan;
   int L = probes[(long long)bq * nprobe + bp];
-  long long out_base = ((long long)bq * nprobe + bp) * k;
-  long long s0 = off[L], s1 = off[L + 1];
+  long long out_base = (((long long)bq * nprobe + bp) * fan + seg) * k;
+  long long l0 = off[L], l1 = off[L + 1];
+  long long len = l1 - l0;
+  long long s0 = l0 + len * seg / fan;
+  long long s1 = l0 + len * (seg + 1) / fan;
   if (s0 == s1) {
     for (int j = threadIdx.x; j < k; j += blockDim.x) {
       cand_d[out_base + j] = DFANN_FLT_MAX;
@@ -1006,11 +1015,12 @@ __device__ void ivf_scan_body(
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
       const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
       int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
-      unsigned *cand_p, int fam_floats) {                                      \
+      unsigned *cand_p, int fam_floats, int fan) {                             \
     ivf_scan_body<FAM, IS_IP, REGSEL>(q, cent, cb, sq_vmin, sq_scale, probes,  \
                                       keys, codes, off, nq, nprobe, d, m,      \
                                       dsub, k, stride, cand_d, cand_p,         \
-                                      fam_floats);                             \
+                                      fam_floats, nullptr, nullptr, nullptr,   \
+                                      fan);                                    \
   }
 
 #define INSTANTIATE_SCAN_PRE(NAME, REGSEL)                                     \
