@@ -193,7 +193,8 @@ struct dfann_index {
                     // shrink it to force the multi-chunk paths)
   int max_ppc = 256;  // k-means subsample cap per centroid (spec "max_ppc")
   bool coarse_bf16 = false;
-  bool pq_pre = false;  // spec "pq_precomputed": PQ-L2 term2/term3 tables  // spec "coarse_bf16": assign/coarse GEMMs on
+  bool pq_pre = false;  // spec "pq_precomputed": PQ-L2 term2/term3 tables
+  int scan_fan = 1;     // spec "scan_fan": list-segment fan (experiment)  // spec "coarse_bf16": assign/coarse GEMMs on
                              // bf16 MFMA (~16x f32 rate) — approximate
                              // ranking path for huge nlist
 
@@ -480,6 +481,9 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->coarse_bf16 = json_int(js, "coarse_bf16", 0) != 0;
   h->max_ppc = (int)json_int(js, "max_ppc", 256);
   h->pq_pre = json_int(js, "pq_precomputed", 0) != 0;
+  h->scan_fan = (int)json_int(js, "scan_fan", 1);
+  if (h->scan_fan < 1) h->scan_fan = 1;
+  if (h->scan_fan > 16) h->scan_fan = 16;
   if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
   if (h->type != T_FLAT && h->nlist <= 0) {
     delete h;
@@ -764,13 +768,10 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   bool use_pre = h->type == T_IVFPQ && h->metric == M_L2 && h->pq_pre &&
                  h->term2.p;
   if (use_pre) fam_floats = h->m * 256;  // LUT only, no rbuf
-  // segment fan: fill the chip when nq*nprobe is small. PQ blocks pay a
-  // per-block LUT build, so only the cheap-staging families fan out.
-  int fan = 1;
-  if (h->type != T_IVFPQ) {
-    int64_t blocks = nq * (int64_t)nprobe;
-    while (fan < 8 && blocks * fan < 65536) fan *= 2;
-  }
+  // segment fan (spec "scan_fan"): kept as an experiment knob — measured
+  // NEGATIVE at the 10M SQ8 shape (per-block staging/extraction overhead
+  // outweighs tail imbalance: 1936 -> 1339 GB/s at fan 8), so default 1.
+  int fan = h->type != T_IVFPQ ? h->scan_fan : 1;
   size_t lds = (size_t)fam_floats * 4 + (rk ? REGSEL_LDS_BYTES : SEL_LDS_BYTES);
   if (lds > 160 * 1024)
     throw std::runtime_error("scan LDS over budget (m too large)");
