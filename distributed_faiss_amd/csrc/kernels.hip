@@ -452,6 +452,108 @@ extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_nt(
 #undef GB_MFMA
 }
 
+// glds-staged variant (cdna_hip_programming.md §5 ladder step 3 /
+// Guideline 15): global->LDS DMA (16-B wide) stages the next tile while
+// the MFMAs run; ONE __shared__ object (a second one de-pipelines glds —
+// §5 ".s-level traps" (a)); LDS image is lane-linear, so the bank
+// swizzle lives on the SOURCE address and the fragment-read offset
+// (rule 21): byte ^= (row&7)<<4 cuts the 16-lane ds_read_b128 group from
+// 8-way to 2-way. Requires K % 8 == 0 (host falls back otherwise).
+extern "C" __global__ __launch_bounds__(256) void k_gemm_bf16_glds(
+    const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
+    const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
+  // [buf][operand][row][col] bf16, linear (swizzle on source/read)
+  __shared__ unsigned short smem2[2][2][GB_T][GB_K];
+  int bi = blockIdx.y * GB_T;
+  int bj = blockIdx.x * GB_T;
+  int tid = threadIdx.x;
+  int lane = tid & 63, w = tid >> 6;
+  int wr = (w >> 1) * 64, wc = (w & 1) * 64;
+  f32x4 acc[4][4] = {};
+  int li = lane & 15;
+  int ke = (lane >> 4) * 8;
+
+  // wave w stages rows [w*32, w*32+32) of each operand: 4 glds per operand,
+  // each covering 64 consecutive 16-B groups (8 rows)
+#define GLDS_STAGE(BUF, K0)                                                    \
+  _Pragma("unroll") for (int i = 0; i < 4; ++i) {                              \
+    int g = (w * 4 + i) * 64 + lane; /* group index, lane-linear */            \
+    int r = g >> 3;                                                            \
+    int c8 = (g & 7) * 8;                                                      \
+    int c8s = c8 ^ ((r & 7) << 3); /* source-side XOR swizzle (elements) */    \
+    unsigned short *ldst =                                                     \
+        &smem2[BUF][0][0][0] + ((size_t)(w * 4 + i) * 64) * 8;                 \
+    const unsigned short *ga = &A[(size_t)(bi + r) * lda + (K0) + c8s];        \
+    const unsigned short *gb = &B[(size_t)(bj + r) * ldb + (K0) + c8s];        \
+    bool oka = (bi + r < M) && ((K0) + c8s + 7 < K);                           \
+    bool okb = (bj + r < N) && ((K0) + c8s + 7 < K);                           \
+    if (oka)                                                                   \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) unsigned int *)ga,          \
+          (__attribute__((address_space(3))) unsigned int *)ldst, 16, 0, 0);   \
+    else                                                                       \
+      *reinterpret_cast<uint4 *>(                                              \
+          &smem2[BUF][0][0][0] + (size_t)g * 8) = uint4{0, 0, 0, 0};           \
+    unsigned short *ldstB =                                                    \
+        &smem2[BUF][1][0][0] + ((size_t)(w * 4 + i) * 64) * 8;                 \
+    if (okb)                                                                   \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) unsigned int *)gb,          \
+          (__attribute__((address_space(3))) unsigned int *)ldstB, 16, 0, 0);  \
+    else                                                                       \
+      *reinterpret_cast<uint4 *>(                                              \
+          &smem2[BUF][1][0][0] + (size_t)g * 8) = uint4{0, 0, 0, 0};           \
+  }
+
+#define GLDS_MFMA(BUF)                                                         \
+  _Pragma("unroll") for (int kk = 0; kk < GB_K; kk += 32) {                    \
+    _Pragma("unroll") for (int ti = 0; ti < 4; ++ti) {                         \
+      int ra = wr + ti * 16 + li;                                              \
+      int ca = (kk + ke) ^ ((ra & 7) << 3);                                    \
+      bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&smem2[BUF][0][ra][ca]);   \
+      _Pragma("unroll") for (int tj = 0; tj < 4; ++tj) {                       \
+        int rb = wc + tj * 16 + li;                                            \
+        int cbx = (kk + ke) ^ ((rb & 7) << 3);                                 \
+        bf16x8 b0 =                                                            \
+            *reinterpret_cast<const bf16x8 *>(&smem2[BUF][1][rb][cbx]);        \
+        acc[ti][tj] =                                                          \
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj],       \
+                                                    0, 0, 0);                  \
+      }                                                                        \
+    }                                                                          \
+  }
+
+  GLDS_STAGE(0, 0)
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = GB_K; k0 < K; k0 += GB_K) {
+    GLDS_STAGE(cur ^ 1, k0)  // DMA in flight during the MFMAs
+    GLDS_MFMA(cur)
+    __syncthreads();  // drains the glds queue (vmcnt(0) inside)
+    cur ^= 1;
+  }
+  GLDS_MFMA(cur)
+
+  int rrow = (lane >> 4) * 4;
+#pragma unroll
+  for (int ti = 0; ti < 4; ++ti) {
+#pragma unroll
+    for (int tj = 0; tj < 4; ++tj) {
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        int row = bi + wr + ti * 16 + rrow + rg;
+        int col = bj + wc + tj * 16 + li;
+        if (row < M && col < N)
+          C[(size_t)row * ldc + col] =
+              gemm_key(acc[ti][tj][rg], row, col, qn, bn, mode);
+      }
+    }
+  }
+#undef GLDS_STAGE
+#undef GLDS_MFMA
+}
+
 // ---------------------------------------------------------------------------
 // row norms ||x_i||^2, one block per row
 // ---------------------------------------------------------------------------
