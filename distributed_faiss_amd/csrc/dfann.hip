@@ -129,7 +129,7 @@ struct DevBuf {
     cap = nb;
   }
   void free() {
-    if (p) hipFree(p);
+    if (p) (void)hipFree(p);  // best-effort in destructor paths
     p = nullptr;
     cap = 0;
   }
@@ -221,8 +221,8 @@ struct dfann_index {
   ~dfann_index() {
     for (auto &v : {ev_scan, ev_gemm, ev_merge})
       for (auto &e : v) {
-        hipEventDestroy(e.a);
-        hipEventDestroy(e.b);
+        (void)hipEventDestroy(e.a);
+        (void)hipEventDestroy(e.b);
       }
   }
 
@@ -1299,8 +1299,8 @@ static double sum_events(std::vector<TimingEv> &v) {
     float el = 0;
     HIP_CHECK(hipEventElapsedTime(&el, e.a, e.b));
     ms += el;
-    hipEventDestroy(e.a);
-    hipEventDestroy(e.b);
+    (void)hipEventDestroy(e.a);
+    (void)hipEventDestroy(e.b);
   }
   v.clear();
   return ms;
