@@ -156,7 +156,7 @@ def main():
         init_from_env,
         merge_gathered,
     )
-    from distributed_faiss_amd.hip_engine import HipEngine, merge_topk_dev
+    from distributed_faiss_amd.hip_engine import HipEngine
 
     rank, world = init_from_env()
     assert world == args.gpus or world == 1, (world, args.gpus)
