@@ -961,42 +961,6 @@ __device__ void ivf_scan_body(
         }
       }
     }
-  } else if (FAM == 0 && m % 4 == 0) {
-    // PQ: 4 lanes per row, lane l sums LUT entries for its quarter of the
-    // subspaces (j in [l*m/4, (l+1)*m/4), ascending), then a fixed 2-step
-    // butterfly: dist = ((p0+p2) + (p1+p3)). Mirrored op-for-op by the
-    // oracle (adc_scan_tree4) — the shared fp32 order of the bit-exact
-    // tier. Cuts the per-row serial LDS-gather chain 4x vs one lane per
-    // row. m % 4 != 0 falls through to the sequential path below.
-    {
-      const float *lut = fam;
-      int g4 = threadIdx.x & 3, grp = threadIdx.x >> 2;  // 64 row-groups
-      int J = m >> 2;                                     // subspaces/lane
-      for (long long base = s0; base < s1; base += 256) {
-        if (!REGSEL) sel_guard(s, k, 256);
-#pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          long long pos = base + (long long)u * 64 + grp;
-          bool valid = pos < s1;
-          float part = 0.f;
-          if (valid) {
-            const uint8_t *cp = codes + pos * (size_t)stride + g4 * J;
-            for (int jj = 0; jj < J; ++jj) {
-#pragma clang fp contract(off)
-              unsigned c = cp[jj];
-              part = part + lut[(g4 * J + jj) * 256 + c];
-            }
-          }
-          part += __shfl_xor(part, 2, 4);
-          part += __shfl_xor(part, 1, 4);
-          float dist = IS_IP ? -(bias + part) : (PRE ? bias + part : part);
-          if (valid && g4 == 0) {
-            if (REGSEL) loc.push(dist, (unsigned)pos);
-            else sel_try(s, dist, (unsigned)pos);
-          }
-        }
-      }
-    }
   } else if (FAM == 2) {
     // SQ8: 8 lanes per row — a wave reads 8 consecutive rows as 1 KiB of
     // CONTIGUOUS lines (the streaming-copy access pattern) instead of 64

@@ -197,27 +197,14 @@ def seq_ip(q: np.ndarray, xs: np.ndarray) -> np.ndarray:
 
 
 def adc_scan(lut: np.ndarray, codes: np.ndarray) -> np.ndarray:
-    """ADC distances, op-for-op the HIP ivfpq scan kernel's fp32 order.
+    """ADC distances: sum_j lut[j, codes[i, j]], sequential over j.
 
-    lut: (m, 256) fp32; codes: (n, m) uint8. For m % 4 == 0 (the kernel's
-    4-lane-per-row path): four partials, each sequential over its quarter
-    of the subspaces, combined by the fixed butterfly
-    ((p0+p2) + (p1+p3)). Otherwise: plain sequential over j.
+    lut: (m, 256) fp32; codes: (n, m) uint8. fp32 accumulation, j ascending
+    — mirrored exactly by the HIP ivfpq scan kernel.
     """
-    m = lut.shape[0]
     n = codes.shape[0]
-    if m % 4 == 0:
-        J = m // 4
-        parts = []
-        for lane in range(4):
-            acc = np.zeros(n, dtype=np.float32)
-            for jj in range(J):
-                j = lane * J + jj
-                acc = acc + lut[j, codes[:, j]]
-            parts.append(acc)
-        return (parts[0] + parts[2]) + (parts[1] + parts[3])
     acc = np.zeros(n, dtype=np.float32)
-    for j in range(m):
+    for j in range(lut.shape[0]):
         acc = acc + lut[j, codes[:, j]]
     return acc
 
