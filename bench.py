@@ -106,13 +106,17 @@ def gen_shard(cfg, rank, device):
     lbl = torch.randint(0, cfg["centers"], (n,), generator=g, device=device)
     xb = centers[lbl] + cfg["sigma"] * torch.randn(n, d, generator=g,
                                                    device=device)
+    # HELD-OUT queries: fresh samples from the SAME
+    # mixture (identical center set on every rank via the shared seed),
+    # not perturbed DB rows — a query's true NN is then a nearby stored
+    # point, frequently near a list boundary, so the nprobe/recall
+    # trade-off is real rather than saturating at nprobe=1.
     gq = torch.Generator(device=device).manual_seed(9999)  # same on all ranks
-    qidx = torch.randint(0, n, (cfg["nq"],), generator=gq, device=device)
-    # queries perturb rank-0's shard points; every rank must use the SAME
-    # queries, so regenerate rank-0 data stats via broadcast-free trick:
-    # queries are only well-defined from rank 0's xb — ranks with rank>0
-    # get them via the same seed when N==1, else from a broadcast.
-    return xb.float().contiguous(), qidx
+    qlbl = torch.randint(0, cfg["centers"], (cfg["nq"],), generator=gq,
+                         device=device)
+    q = centers[qlbl] + cfg["sigma"] * torch.randn(
+        cfg["nq"], d, generator=gq, device=device)
+    return xb.float().contiguous(), q.float().contiguous()
 
 
 def exact_ground_truth(xb, q, metric, k, device):
@@ -169,20 +173,14 @@ def main():
 
     # ---- build phase (untimed) ----
     t0 = time.time()
-    xb, qidx = gen_shard(cfg, rank, device)
-    # queries: identical across ranks — derived from rank 0's shard
+    xb, q = gen_shard(cfg, rank, device)
+    # NB queries are seeded identically on every rank (gen_shard), but the
+    # CENTERS differ per rank (each shard its own mixture) — queries come
+    # from rank 0's centers, so broadcast to keep them truly identical
     if world > 1:
         import torch.distributed as dist
 
-        # queries derive from rank 0's shard; broadcast only the selected rows
-        qrows = xb[qidx] if rank == 0 else torch.empty(
-            cfg["nq"], cfg["d"], device=device)
-        dist.broadcast(qrows, src=0)
-    else:
-        qrows = xb[qidx]
-    gq = torch.Generator(device=device).manual_seed(9999 + 7)
-    q = (qrows + 0.1 * cfg["sigma"] * torch.randn(
-        cfg["nq"], cfg["d"], generator=gq, device=device)).contiguous()
+        dist.broadcast(q, src=0)
     log(f"data generated in {time.time()-t0:.1f}s")
 
     spec = {"type": cfg["type"], "dim": cfg["d"], "metric": metric,
