@@ -886,13 +886,40 @@ __device__ void ivf_scan_body(
       const float *cbe = cb + ((size_t)j * 256 + c) * dsub;
       const float *rs = rbuf + j * dsub;
       float acc = 0.f;
-      for (int t = 0; t < dsub; ++t) {
+      if ((dsub & 3) == 0 && dsub <= 32) {
+        // float4 codebook loads (up to 8 in flight); accumulation order
+        // unchanged (t ascending, mul+add, contract off) — bit-exact
+        float4 cv[8];
+        int nv = dsub >> 2;
+#pragma unroll
+        for (int v = 0; v < 8; ++v)
+          if (v < nv) cv[v] = reinterpret_cast<const float4 *>(cbe)[v];
+#pragma unroll
+        for (int v = 0; v < 8; ++v) {
+          if (v < nv) {
 #pragma clang fp contract(off)
-        if (IS_IP) {
-          acc = acc + rs[t] * cbe[t];
-        } else {
-          float diff = rs[t] - cbe[t];
-          acc = acc + diff * diff;
+            for (int tt = 0; tt < 4; ++tt) {
+              float cbv = tt == 0 ? cv[v].x : tt == 1 ? cv[v].y
+                          : tt == 2 ? cv[v].z : cv[v].w;
+              int t = v * 4 + tt;
+              if (IS_IP) {
+                acc = acc + rs[t] * cbv;
+              } else {
+                float diff = rs[t] - cbv;
+                acc = acc + diff * diff;
+              }
+            }
+          }
+        }
+      } else {
+        for (int t = 0; t < dsub; ++t) {
+#pragma clang fp contract(off)
+          if (IS_IP) {
+            acc = acc + rs[t] * cbe[t];
+          } else {
+            float diff = rs[t] - cbe[t];
+            acc = acc + diff * diff;
+          }
         }
       }
       lut[e] = acc;
