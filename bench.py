@@ -162,6 +162,8 @@ def main():
     ap.add_argument("--target-recall", type=float, default=0.95)
     ap.add_argument("--cpu-baseline", type=int, default=1)
     ap.add_argument("--cpu-sample-queries", type=int, default=256)
+    ap.add_argument("--pq-precomputed", type=int, default=-1,
+                    help="-1: workload default; 0/1 override")
     args = ap.parse_args()
 
     import torch
@@ -200,7 +202,9 @@ def main():
             "coarse_bf16": cfg.get("coarse_bf16", 0),
             "max_ppc": cfg.get("max_ppc", 256),
             "ws_mb": cfg.get("ws_mb", 512),
-            "pq_precomputed": cfg.get("pq_precomputed", 0)}
+            "pq_precomputed": (cfg.get("pq_precomputed", 0)
+                               if args.pq_precomputed < 0
+                               else args.pq_precomputed)}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)
