@@ -776,6 +776,10 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   size_t lds = (size_t)fam_floats * 4 + (rk ? REGSEL_LDS_BYTES : SEL_LDS_BYTES);
   if (lds > 160 * 1024)
     throw std::runtime_error("scan LDS over budget (m too large)");
+  // big-LDS blocks (m=64 LUTs: ~69 KB -> 2 blocks/CU) run 512 threads so
+  // the CU still holds 4 waves/SIMD (register path only; the LDS-buffer
+  // selection path is capacity-sized for 256)
+  unsigned scan_bs = (rk && lds >= 32 * 1024) ? 512 : 256;
   h->ws3.ensure((size_t)nq * nprobe * fan * k * 4);
   h->ws4.ensure((size_t)nq * nprobe * fan * k * 4);
   cand_d = h->ws3.as<float>();
@@ -791,7 +795,7 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        0, stream, q, h->codebooks.as<float>(), nq, h->m,
                        h->dsub, h->term3_ws.as<float>());
     auto pk = rk ? k_scan_pq_l2_pre_rk : k_scan_pq_l2_pre;
-    hipLaunchKernelGGL(pk, dim3((unsigned)(nq * nprobe)), dim3(256), lds,
+    hipLaunchKernelGGL(pk, dim3((unsigned)(nq * nprobe)), dim3(scan_bs), lds,
                        stream, q, h->centroids.as<float>(),
                        h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                        h->sq_scale.as<float>(), probes, keys,
@@ -800,8 +804,8 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        cand_d, cand_p, fam_floats, h->term2.as<float>(),
                        h->term3_ws.as<float>(), h->qn_ws.as<float>());
   } else {
-    hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe * fan)), dim3(256),
-                       lds, stream, q, h->centroids.as<float>(),
+    hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe * fan)),
+                       dim3(scan_bs), lds, stream, q, h->centroids.as<float>(),
                        h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                        h->sq_scale.as<float>(), probes, keys,
                        h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),

@@ -171,15 +171,16 @@ struct RegTopK {
 
 // Block-wide merge of per-thread RegTopK heads. Phase 1: each wave
 // extracts its own top-k with pure shfl rounds (no barriers); phase 2:
-// one barrier, then thread 0 4-way-merges the per-wave sorted lists.
-// lds: 4*16*8 = 512 B scratch. Requires blockDim == 256 (4 waves).
-#define REGSEL_LDS_BYTES 640
+// one barrier, then thread 0 merges the per-wave sorted lists.
+// lds: 8*16*8 = 1 KiB scratch. Supports blockDim 256 or 512 (4/8 waves).
+#define REGSEL_LDS_BYTES 1152
 template <int K>
 __device__ void regtopk_block_extract(RegTopK<K> &loc, int k, char *lds,
                                       float *out_d, unsigned *out_p) {
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  float *wvd = reinterpret_cast<float *>(lds);             // [4][K]
-  unsigned *wvp = reinterpret_cast<unsigned *>(lds + 4 * K * 4);
+  int nw = blockDim.x >> 6;  // 4 or 8 waves
+  float *wvd = reinterpret_cast<float *>(lds);             // [8][K]
+  unsigned *wvp = reinterpret_cast<unsigned *>(lds + 8 * K * 4);
   for (int round = 0; round < k; ++round) {
     float cd = loc.d[0];
     unsigned cp = loc.p[0];
@@ -202,29 +203,25 @@ __device__ void regtopk_block_extract(RegTopK<K> &loc, int k, char *lds,
   }
   __syncthreads();
   if (threadIdx.x == 0) {
-    int c0 = 0, c1 = 0, c2 = 0, c3 = 0;
+    int cur[8] = {};
     for (int round = 0; round < k; ++round) {
-      // a wave list holds exactly k entries; an exhausted list reads as pad
-      float b0 = c0 < k ? wvd[0 * K + c0] : DFANN_FLT_MAX;
-      float b1 = c1 < k ? wvd[1 * K + c1] : DFANN_FLT_MAX;
-      float b2 = c2 < k ? wvd[2 * K + c2] : DFANN_FLT_MAX;
-      float b3 = c3 < k ? wvd[3 * K + c3] : DFANN_FLT_MAX;
-      unsigned p0 = c0 < k ? wvp[0 * K + c0] : PAD_POS;
-      unsigned p1 = c1 < k ? wvp[1 * K + c1] : PAD_POS;
-      unsigned p2 = c2 < k ? wvp[2 * K + c2] : PAD_POS;
-      unsigned p3 = c3 < k ? wvp[3 * K + c3] : PAD_POS;
       int bw = 0;
-      float bd = b0;
-      unsigned bp = p0;
-      if (sel_less(b1, p1, bd, bp)) { bd = b1; bp = p1; bw = 1; }
-      if (sel_less(b2, p2, bd, bp)) { bd = b2; bp = p2; bw = 2; }
-      if (sel_less(b3, p3, bd, bp)) { bd = b3; bp = p3; bw = 3; }
+      float bd = DFANN_FLT_MAX;
+      unsigned bp = PAD_POS;
+#pragma unroll
+      for (int v = 0; v < 8; ++v) {
+        if (v < nw) {
+          // a wave list holds exactly k entries; exhausted reads as pad
+          float bv = cur[v] < k ? wvd[v * K + cur[v]] : DFANN_FLT_MAX;
+          unsigned pv = cur[v] < k ? wvp[v * K + cur[v]] : PAD_POS;
+          if (v == 0 || sel_less(bv, pv, bd, bp)) { bd = bv; bp = pv; bw = v; }
+        }
+      }
       out_d[round] = bd;
       out_p[round] = bp;
-      if (bw == 0) ++c0;
-      else if (bw == 1) ++c1;
-      else if (bw == 2) ++c2;
-      else ++c3;
+#pragma unroll
+      for (int v = 0; v < 8; ++v)
+        if (v == bw) ++cur[v];  // static index (rule 20: no scratch)
     }
   }
   __syncthreads();
@@ -593,11 +590,12 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows(
   sel_init(s);
   __syncthreads();
   const float *kp = keys + row * ldk;
-  for (long long c0 = 0; c0 < cols; c0 += 512) {
-    sel_guard(s, k, 512);
+  const int BS = blockDim.x;
+  for (long long c0 = 0; c0 < cols; c0 += (long long)2 * BS) {
+    sel_guard(s, k, 2 * BS);
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
-      long long c = c0 + u * 256 + threadIdx.x;
+      long long c = c0 + u * BS + threadIdx.x;
       if (c < cols) sel_try(s, kp[c], (unsigned)c + base);
     }
   }
@@ -960,11 +958,12 @@ __device__ void ivf_scan_body(
   // --- scan ---
   if (FAM == 1) {
     // 16-lane-per-vector tree reduction (tolerance parity path)
-    int sub = threadIdx.x & 15, grp = threadIdx.x >> 4;  // 16 groups
-    for (long long base = s0; base < s1; base += 16 * 8) {
-      if (!REGSEL) sel_guard(s, k, 128);
+    int sub = threadIdx.x & 15, grp = threadIdx.x >> 4;
+    const int NG16 = blockDim.x >> 4;  // row groups per block
+    for (long long base = s0; base < s1; base += (long long)NG16 * 8) {
+      if (!REGSEL) sel_guard(s, k, NG16 * 8);
       for (int u = 0; u < 8; ++u) {
-        long long pos = base + (long long)u * 16 + grp;
+        long long pos = base + (long long)u * NG16 + grp;
         float dist = 0.f;
         bool valid = pos < s1;
         if (valid) {
@@ -998,8 +997,9 @@ __device__ void ivf_scan_body(
       const float *ubuf = fam, *vbuf = fam + d;
       int g8 = threadIdx.x & 7, grp = threadIdx.x >> 3;  // 32 row-groups
       const bool onechunk = d <= 128;  // lane covers one 16-B chunk
-      for (long long base = s0; base < s1; base += 128) {
-        if (!REGSEL) sel_guard(s, k, 128);
+      const int NG8 = blockDim.x >> 3;  // row groups per block
+      for (long long base = s0; base < s1; base += (long long)NG8 * 4) {
+        if (!REGSEL) sel_guard(s, k, NG8 * 4);
         if (onechunk) {
           // issue all 4 row-set loads up front: 4 independent HBM
           // requests in flight per wave instead of 1 (latency hiding)
@@ -1007,7 +1007,7 @@ __device__ void ivf_scan_body(
           bool val4[4];
 #pragma unroll
           for (int u = 0; u < 4; ++u) {
-            long long pos = base + (long long)u * 32 + grp;
+            long long pos = base + (long long)u * NG8 + grp;
             val4[u] = pos < s1;
             int t0 = g8 * 16;
             wv4[u] = (val4[u] && t0 < d)
@@ -1017,7 +1017,7 @@ __device__ void ivf_scan_body(
           }
 #pragma unroll
           for (int u = 0; u < 4; ++u) {
-            long long pos = base + (long long)u * 32 + grp;
+            long long pos = base + (long long)u * NG8 + grp;
             float part = 0.f;
             int t0 = g8 * 16;
             if (t0 < d) {
@@ -1051,7 +1051,7 @@ __device__ void ivf_scan_body(
         } else {
 #pragma unroll
           for (int u = 0; u < 4; ++u) {
-            long long pos = base + (long long)u * 32 + grp;
+            long long pos = base + (long long)u * NG8 + grp;
             bool valid = pos < s1;
             float part = 0.f;
             if (valid) {
@@ -1095,11 +1095,12 @@ __device__ void ivf_scan_body(
     // full cache line each (8 uint4 in registers), so 2 in flight keeps
     // VGPRs ~100 (5 waves/SIMD) instead of 161 (3 waves).
     const int UROWS = (FAM == 0) ? 4 : 2;
-    for (long long base = s0; base < s1; base += (long long)UROWS * 256) {
+    const int BS = blockDim.x;
+    for (long long base = s0; base < s1; base += (long long)UROWS * BS) {
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         if (u >= UROWS) break;
-        long long pos = base + (long long)u * 256 + threadIdx.x;
+        long long pos = base + (long long)u * BS + threadIdx.x;
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
           float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
@@ -1109,11 +1110,12 @@ __device__ void ivf_scan_body(
       }
     }
   } else {
-    for (long long base = s0; base < s1; base += 512) {
-      sel_guard(s, k, 512);
+    const int BS = blockDim.x;
+    for (long long base = s0; base < s1; base += (long long)2 * BS) {
+      sel_guard(s, k, 2 * BS);
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
-        long long pos = base + (long long)u * 256 + threadIdx.x;
+        long long pos = base + (long long)u * BS + threadIdx.x;
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
           float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
@@ -1139,7 +1141,7 @@ __device__ void ivf_scan_body(
 }
 
 #define INSTANTIATE_SCAN(NAME, FAM, IS_IP, REGSEL)                             \
-  extern "C" __global__ __launch_bounds__(256) void NAME(                      \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
       const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
@@ -1153,7 +1155,7 @@ __device__ void ivf_scan_body(
   }
 
 #define INSTANTIATE_SCAN_PRE(NAME, REGSEL)                                     \
-  extern "C" __global__ __launch_bounds__(256) void NAME(                      \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
       const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
@@ -1207,11 +1209,12 @@ extern "C" __global__ __launch_bounds__(256) void k_merge_cand(
   __syncthreads();
   const float *cd = cand_d + qi * C;
   const unsigned *cp = cand_p + qi * C;
-  for (int c0 = 0; c0 < C; c0 += 512) {
-    sel_guard(s, k, 512);
+  const int BSm = blockDim.x;
+  for (int c0 = 0; c0 < C; c0 += 2 * BSm) {
+    sel_guard(s, k, 2 * BSm);
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
-      int c = c0 + u * 256 + threadIdx.x;
+      int c = c0 + u * BSm + threadIdx.x;
       if (c < C) {
         unsigned pos = cp[c];
         if (pos != PAD_POS) {
@@ -1255,11 +1258,12 @@ extern "C" __global__ __launch_bounds__(256) void k_merge_shards(
   sel_init(s);
   __syncthreads();
   int C = S * k;
-  for (int c0 = 0; c0 < C; c0 += 512) {
-    sel_guard(s, k, 512);
+  const int BSm = blockDim.x;
+  for (int c0 = 0; c0 < C; c0 += 2 * BSm) {
+    sel_guard(s, k, 2 * BSm);
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
-      int c = c0 + u * 256 + threadIdx.x;
+      int c = c0 + u * BSm + threadIdx.x;
       if (c < C) {
         int sh = c / k, j = c % k;
         float v = Dall[((long long)sh * nq + qi) * k + j];
