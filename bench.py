@@ -62,7 +62,7 @@ WORKLOADS = {
     # 10M single-GPU stress (toward configs[3] scale; 160 MB codes)
     "ivfpq_10m_d128_m16": dict(
         type="ivfpq", d=128, n=10_000_000, nlist=4096, m=16, nbits=8,
-        metric=1, nq=10_000, k=10, centers=50_000, sigma=0.5, latent=12,
+        metric=1, nq=10_000, k=10, centers=200_000, sigma=0.5, latent=12,
     ),
     # true HBM-bound scan: 10M x 128B SQ8 codes = 1.28 GB >> 256 MB L3
     "ivfsq8_10m_d128": dict(
