@@ -44,7 +44,7 @@ WORKLOADS = {
     # non-degenerate (sigma 0.15 was solved at nprobe=1).
     "ivfpq_1m_d128_m16": dict(
         type="ivfpq", d=128, n=1_000_000, nlist=1024, m=16, nbits=8,
-        metric=1, nq=10_000, k=10, centers=10_000, sigma=0.5, latent=12,
+        metric=1, nq=10_000, k=10, centers=10_000, sigma=0.45, latent=12,
     ),
     # BASELINE.json configs[1] (ivf_simple 1M, dot) — parity/regression
     "ivfflat_1m_d128": dict(
