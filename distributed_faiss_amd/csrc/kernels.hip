@@ -884,36 +884,32 @@ __device__ void ivf_scan_body(
       const float *cbe = cb + ((size_t)j * 256 + c) * dsub;
       const float *rs = rbuf + j * dsub;
       float acc = 0.f;
-      // per-dsub specializations with UNCONDITIONAL float4 loads: a
-      // runtime-guarded per-element load makes hipcc branch around each
-      // load and drain vmcnt(0) per element (cdna_hip_programming.md §5
-      // ".s-level traps" (c)) — serialized L2 round trips. Accumulation
-      // order unchanged (t ascending, mul+add, contract off) — bit-exact.
-#define DFANN_LUT_ACC(NV)                                                      \
-      {                                                                        \
-        float4 cv[NV];                                                         \
-        _Pragma("unroll") for (int v = 0; v < (NV); ++v)                       \
-            cv[v] = reinterpret_cast<const float4 *>(cbe)[v];                  \
-        _Pragma("unroll") for (int v = 0; v < (NV); ++v) {                     \
-          _Pragma("clang fp contract(off)")                                    \
-          _Pragma("unroll") for (int tt = 0; tt < 4; ++tt) {                   \
-            float cbv = tt == 0 ? cv[v].x : tt == 1 ? cv[v].y                  \
-                        : tt == 2 ? cv[v].z : cv[v].w;                         \
-            int t = v * 4 + tt;                                                \
-            if (IS_IP) {                                                       \
-              acc = acc + rs[t] * cbv;                                         \
-            } else {                                                           \
-              float diff = rs[t] - cbv;                                        \
-              acc = acc + diff * diff;                                         \
-            }                                                                  \
-          }                                                                    \
-        }                                                                      \
-      }
-      if (dsub == 4) DFANN_LUT_ACC(1)
-      else if (dsub == 8) DFANN_LUT_ACC(2)
-      else if (dsub == 12) DFANN_LUT_ACC(3)
-      else if (dsub == 16) DFANN_LUT_ACC(4)
-      else {
+      if ((dsub & 3) == 0 && dsub <= 32) {
+        // float4 codebook loads (up to 8 in flight); accumulation order
+        // unchanged (t ascending, mul+add, contract off) — bit-exact
+        float4 cv[8];
+        int nv = dsub >> 2;
+#pragma unroll
+        for (int v = 0; v < 8; ++v)
+          if (v < nv) cv[v] = reinterpret_cast<const float4 *>(cbe)[v];
+#pragma unroll
+        for (int v = 0; v < 8; ++v) {
+          if (v < nv) {
+#pragma clang fp contract(off)
+            for (int tt = 0; tt < 4; ++tt) {
+              float cbv = tt == 0 ? cv[v].x : tt == 1 ? cv[v].y
+                          : tt == 2 ? cv[v].z : cv[v].w;
+              int t = v * 4 + tt;
+              if (IS_IP) {
+                acc = acc + rs[t] * cbv;
+              } else {
+                float diff = rs[t] - cbv;
+                acc = acc + diff * diff;
+              }
+            }
+          }
+        }
+      } else {
         for (int t = 0; t < dsub; ++t) {
 #pragma clang fp contract(off)
           if (IS_IP) {
@@ -924,7 +920,6 @@ __device__ void ivf_scan_body(
           }
         }
       }
-#undef DFANN_LUT_ACC
       lut[e] = acc;
     }
   } else if (FAM == 1) {
