@@ -164,6 +164,8 @@ def main():
     ap.add_argument("--cpu-sample-queries", type=int, default=256)
     ap.add_argument("--pq-precomputed", type=int, default=-1,
                     help="-1: workload default; 0/1 override")
+    ap.add_argument("--pq-lut-global", type=int, default=-1,
+                    help="-1: auto (on at m>=32); 0/1 override")
     args = ap.parse_args()
 
     import torch
@@ -204,7 +206,8 @@ def main():
             "ws_mb": cfg.get("ws_mb", 512),
             "pq_precomputed": (cfg.get("pq_precomputed", 0)
                                if args.pq_precomputed < 0
-                               else args.pq_precomputed)}
+                               else args.pq_precomputed),
+            "pq_lut_global": args.pq_lut_global}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)

@@ -194,6 +194,9 @@ struct dfann_index {
   int max_ppc = 256;  // k-means subsample cap per centroid (spec "max_ppc")
   bool coarse_bf16 = false;
   bool pq_pre = false;  // spec "pq_precomputed": PQ-L2 term2/term3 tables
+  int pq_lut_global = -1;  // spec "pq_lut_global": ADC LUTs built to HBM
+                           // by k_pq_lut, scan stages them coalesced
+                           // (-1 auto: on at m >= 32; 0 off; 1 force)
   int scan_fan = 1;     // spec "scan_fan": list-segment fan (experiment)  // spec "coarse_bf16": assign/coarse GEMMs on
                              // bf16 MFMA (~16x f32 rate) — approximate
                              // ranking path for huge nlist
@@ -212,6 +215,7 @@ struct dfann_index {
   DevBuf ws1, ws2, ws3, ws4, ws5, ws_bf16a, ws_bf16b;
   DevBuf cent_bf16;
   DevBuf term2, term3_ws, qn_ws;  // PQ-L2 precomputed tables
+  DevBuf pq_lut_ws;  // HBM ADC LUTs for the GLUT scan path
 
   // timing
   bool timing = false;
@@ -482,6 +486,7 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->coarse_bf16 = json_int(js, "coarse_bf16", 0) != 0;
   h->max_ppc = (int)json_int(js, "max_ppc", 256);
   h->pq_pre = json_int(js, "pq_precomputed", 0) != 0;
+  h->pq_lut_global = (int)json_int(js, "pq_lut_global", -1);
   h->scan_fan = (int)json_int(js, "scan_fan", 1);
   if (h->scan_fan < 1) h->scan_fan = 1;
   if (h->scan_fan > 16) h->scan_fan = 16;
@@ -769,6 +774,16 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   bool use_pre = h->type == T_IVFPQ && h->metric == M_L2 && h->pq_pre &&
                  h->term2.p;
   if (use_pre) fam_floats = h->m * 256;  // LUT only, no rbuf
+  // GLUT path: LUTs built once to HBM by k_pq_lut, scan stages them as
+  // coalesced float4 slabs. Auto at m >= 32: the in-kernel build
+  // re-reads the full m*256*dsub codebook from L2 per (query, probe)
+  // block and dominates the launch (measured 8.1 ms/launch at the
+  // configs[3] shape, ~125 GB/s of algorithmic code bytes).
+  bool use_glut = h->type == T_IVFPQ && !use_pre &&
+                  (h->pq_lut_global == 1 ||
+                   (h->pq_lut_global < 0 && h->m >= 32)) &&
+                  h->dsub <= 64;
+  if (use_glut) fam_floats = h->m * 256;  // LUT only, no rbuf
   // segment fan (spec "scan_fan"): kept as an experiment knob — measured
   // NEGATIVE at the 10M SQ8 shape (per-block staging/extraction overhead
   // outweighs tail imbalance: 1936 -> 1339 GB/s at fan 8), so default 1.
@@ -803,6 +818,38 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
                        cand_d, cand_p, fam_floats, h->term2.as<float>(),
                        h->term3_ws.as<float>(), h->qn_ws.as<float>());
+  } else if (use_glut) {
+    auto gk = rk ? (ip ? k_scan_pq_ip_g_rk : k_scan_pq_l2_g_rk)
+                 : (ip ? k_scan_pq_ip_g : k_scan_pq_l2_g);
+    // chunk queries so the HBM LUT stays inside the workspace budget
+    size_t row_b = (size_t)nprobe * h->m * 1024;  // LUT bytes per query
+    int64_t qch =
+        std::max<int64_t>(1, (int64_t)(((size_t)h->ws_mb << 20) / row_b));
+    if (qch > nq) qch = nq;
+    h->pq_lut_ws.ensure((size_t)qch * row_b);
+    float *lutg = h->pq_lut_ws.as<float>();
+    int pad = h->dsub | 1;
+    size_t lut_lds = (size_t)(256 + PQ_LUT_QPT) * pad * 4;
+    for (int64_t q0 = 0; q0 < nq; q0 += qch) {
+      int64_t nqc = std::min<int64_t>(qch, nq - q0);
+      long long qpn = (long long)nqc * nprobe;
+      dim3 lg((unsigned)((qpn + PQ_LUT_QPT - 1) / PQ_LUT_QPT),
+              (unsigned)h->m);
+      hipLaunchKernelGGL(k_pq_lut, lg, dim3(256), lut_lds, stream,
+                         q + q0 * h->d, h->centroids.as<float>(),
+                         h->codebooks.as<float>(), probes + q0 * nprobe,
+                         (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
+                         lutg);
+      hipLaunchKernelGGL(gk, dim3((unsigned)(nqc * nprobe)), dim3(scan_bs),
+                         lds, stream, q + q0 * h->d, h->centroids.as<float>(),
+                         h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                         h->sq_scale.as<float>(), probes + q0 * nprobe,
+                         keys + q0 * nprobe, h->cr_codes.as<uint8_t>(),
+                         h->cr_off.as<int64_t>(), (int)nqc, nprobe, h->d,
+                         h->m, h->dsub, k, h->stride,
+                         cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
+                         fam_floats, lutg);
+    }
   } else {
     hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe * fan)),
                        dim3(scan_bs), lds, stream, q, h->centroids.as<float>(),

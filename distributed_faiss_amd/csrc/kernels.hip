@@ -821,7 +821,8 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
   return acc;
 }
 
-template <int FAM, bool IS_IP, bool REGSEL, bool PRE = false>
+template <int FAM, bool IS_IP, bool REGSEL, bool PRE = false,
+          bool GLUT = false>
 __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
@@ -832,7 +833,8 @@ __device__ void ivf_scan_body(
     unsigned *__restrict__ cand_p, int fam_floats,
     const float *__restrict__ term2 = nullptr,
     const float *__restrict__ term3 = nullptr,
-    const float *__restrict__ qn = nullptr, int fan = 1) {
+    const float *__restrict__ qn = nullptr, int fan = 1,
+    const float *__restrict__ glut = nullptr) {
   // fan > 1: each (query, probe) list is split into `fan` segments, one
   // block per segment — long-list tail imbalance at low block counts.
   // EXACT results: every segment's top-k contains the segment's global
@@ -872,6 +874,20 @@ __device__ void ivf_scan_body(
     const float *t3 = term3 + (size_t)bq * m * 256;
     for (int e = threadIdx.x; e < m * 256; e += blockDim.x)
       lut[e] = t2[e] - 2.0f * t3[e];
+  } else if (FAM == 0 && GLUT) {
+    // LUT precomputed in HBM by k_pq_lut (one 1-KiB row per (query,
+    // probe, subspace)): stage it with coalesced float4 copies instead
+    // of recomputing it from the codebook — at m=64 the in-kernel build
+    // re-reads the whole 786-KB codebook from L2 per block and is
+    // latency-bound (measured: the scan launch runs at ~125 GB/s of
+    // algorithmic code bytes at the configs[3] shape). Values are
+    // BIT-IDENTICAL to the in-kernel path: k_pq_lut uses the same
+    // sequential-t accumulation with contract off.
+    float4 *dst = reinterpret_cast<float4 *>(fam);
+    const float4 *src = reinterpret_cast<const float4 *>(
+        glut + ((size_t)bq * nprobe + bp) * ((size_t)m * 256));
+    int n4 = m * 64;
+    for (int e = threadIdx.x; e < n4; e += blockDim.x) dst[e] = src[e];
   } else if (FAM == 0) {
     // rbuf (d floats) AFTER the LUT
     float *lut = fam;
@@ -1182,6 +1198,87 @@ INSTANTIATE_SCAN(k_scan_sqf_l2, 3, false, false)
 INSTANTIATE_SCAN(k_scan_sqf_ip, 3, true, false)
 INSTANTIATE_SCAN(k_scan_pq_l2_rk, 0, false, true)
 INSTANTIATE_SCAN(k_scan_pq_ip_rk, 0, true, true)
+
+// GLUT variants: LUT staged from HBM (built by k_pq_lut) instead of
+// computed in the scan prologue — engine auto-selects at large m where
+// the in-kernel build's per-block codebook re-reads dominate.
+#define INSTANTIATE_SCAN_GLUT(NAME, IS_IP, REGSEL)                             \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
+      unsigned *cand_p, int fam_floats, const float *glut) {                   \
+    ivf_scan_body<0, IS_IP, REGSEL, false, true>(                              \
+        q, cent, cb, sq_vmin, sq_scale, probes, keys, codes, off, nq, nprobe,  \
+        d, m, dsub, k, stride, cand_d, cand_p, fam_floats, nullptr, nullptr,   \
+        nullptr, 1, glut);                                                     \
+  }
+INSTANTIATE_SCAN_GLUT(k_scan_pq_l2_g, false, false)
+INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g, true, false)
+INSTANTIATE_SCAN_GLUT(k_scan_pq_l2_g_rk, false, true)
+INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g_rk, true, true)
+
+// ---------------------------------------------------------------------------
+// k_pq_lut: ADC lookup tables to HBM, one 256-entry row per (query,
+// probe, subspace) at out[(qp)*m*256 + j*256 + c] — the scan's GLUT
+// staging then reads its (query, probe) block as one contiguous,
+// perfectly-coalesced m-KiB slab. grid = (qp tiles, m subspaces),
+// 256 threads = one code c each. The subspace codebook is staged ONCE
+// per block in LDS (odd row stride -> conflict-free reads), so the
+// codebook leaves L2 ~QPT times less often than the in-kernel build.
+// Residual and accumulation are op-for-op the ones the in-kernel build
+// uses (sequential t, mul+add, contract off) -> bit-identical LUTs,
+// oracle parity preserved (oracle/core.py adc_scan).
+// ---------------------------------------------------------------------------
+#define PQ_LUT_QPT 64
+extern "C" __global__ __launch_bounds__(256) void k_pq_lut(
+    const float *__restrict__ q, const float *__restrict__ cent,
+    const float *__restrict__ cb, const int *__restrict__ probes, int nq,
+    int nprobe, int d, int m, int dsub, int is_ip, float *__restrict__ out) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  int pad = dsub | 1;  // odd stride -> gcd(pad, banks) == 1
+  float *cb_sm = reinterpret_cast<float *>(smem);  // 256 * pad
+  float *r_sm = cb_sm + 256 * pad;                 // PQ_LUT_QPT * pad
+  int j = blockIdx.y;
+  const float *cbj = cb + (size_t)j * 256 * dsub;
+  for (int e = threadIdx.x; e < 256 * dsub; e += blockDim.x)
+    cb_sm[(e / dsub) * pad + (e % dsub)] = cbj[e];
+  long long qpn = (long long)nq * nprobe;
+  long long qp0 = (long long)blockIdx.x * PQ_LUT_QPT;
+  long long qp1 = qp0 + PQ_LUT_QPT;
+  if (qp1 > qpn) qp1 = qpn;
+  int nrow = (int)(qp1 - qp0);
+  for (int e = threadIdx.x; e < nrow * dsub; e += blockDim.x) {
+    int rr = e / dsub, t = e % dsub;
+    long long qp = qp0 + rr;
+    int bq = (int)(qp / nprobe);
+    float qv = q[(size_t)bq * d + (size_t)j * dsub + t];
+    if (is_ip) {
+      r_sm[rr * pad + t] = qv;
+    } else {
+      int L = probes[qp];
+      r_sm[rr * pad + t] = qv - cent[(size_t)L * d + (size_t)j * dsub + t];
+    }
+  }
+  __syncthreads();
+  int c = threadIdx.x;
+  const float *crow = cb_sm + c * pad;
+  for (int rr = 0; rr < nrow; ++rr) {
+    const float *rs = r_sm + rr * pad;
+    float acc = 0.f;
+    for (int t = 0; t < dsub; ++t) {
+#pragma clang fp contract(off)
+      if (is_ip) {
+        acc = acc + rs[t] * crow[t];
+      } else {
+        float diff = rs[t] - crow[t];
+        acc = acc + diff * diff;
+      }
+    }
+    out[(size_t)(qp0 + rr) * ((size_t)m * 256) + (size_t)j * 256 + c] = acc;
+  }
+}
 INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
 INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)
 INSTANTIATE_SCAN(k_scan_sq8_l2_rk, 2, false, true)

@@ -340,6 +340,57 @@ def test_pq_precomputed_table_path():
     np.testing.assert_allclose(D0[same], D1[same], rtol=2e-3, atol=2e-3)
 
 
+@pytest.mark.parametrize("metric", [IP, L2])
+@pytest.mark.parametrize("m", [8, 32])
+def test_pq_glut_path_bitexact(m, metric):
+    # HBM-LUT scan path (k_pq_lut + GLUT staging) vs the in-kernel LUT
+    # build AND vs the oracle: k_pq_lut uses the identical sequential-t
+    # accumulation, so all three must agree BITWISE. m=32 also covers
+    # the auto-gate (pq_lut_global=-1 turns the path on at m >= 32);
+    # ws_mb=1 at m=8 forces the query-chunk loop (multiple k_pq_lut +
+    # scan launches per search).
+    d, nlist, per = 64, 8, 300
+    cent, xb = _clustered(nlist, per, d, seed=41)
+    rng = np.random.default_rng(42)
+    cb = rng.standard_normal((m, 256, d // m)).astype(np.float32)
+    lbl = rng.integers(0, nlist, xb.shape[0])
+    cw = rng.integers(0, 256, (xb.shape[0], m))
+    dec = np.concatenate([cb[j][cw[:, j]] for j in range(m)], axis=1)
+    xb = (cent[lbl] + dec
+          + 1e-3 * rng.standard_normal(xb.shape)).astype(np.float32)
+    q = xb[::17][:30] + 0.01 * rng.standard_normal((30, d)).astype(np.float32)
+    q = q.astype(np.float32)
+    res = []
+    for glut, ws in ((0, 512), (1, 512), (1, 1)):
+        spec = {"type": "ivfpq", "dim": d, "metric": metric, "nlist": nlist,
+                "m": m, "nprobe": nlist, "seed": 7, "pq_lut_global": glut,
+                "ws_mb": ws}
+        eng = HipEngine(spec=spec)
+        eng.set_trained(cent, cb)
+        eng.add(xb)
+        probes, keys = eng.coarse(q, nlist)
+        res.append(eng.search_preassigned(q, probes, keys, 10))
+    (D0, I0), (D1, I1), (D2, I2) = res
+    np.testing.assert_array_equal(I0, I1)
+    np.testing.assert_array_equal(D0, D1)  # BITWISE vs in-kernel build
+    np.testing.assert_array_equal(I1, I2)
+    np.testing.assert_array_equal(D1, D2)  # BITWISE across chunk sizes
+    spec_o = {"type": "ivfpq", "dim": d, "metric": metric, "nlist": nlist,
+              "m": m, "nprobe": nlist, "seed": 7}
+    orc = make_oracle_engine(spec_o)
+    orc.centroids, orc.codebooks, orc.is_trained = cent, cb, True
+    orc.add(xb)
+    orc.nprobe = nlist
+    eng = HipEngine(spec=dict(spec_o, pq_lut_global=1))
+    eng.set_trained(cent, cb)
+    eng.add(xb)
+    probes, keys = eng.coarse(q, nlist)
+    D, I = eng.search_preassigned(q, probes, keys, 10)
+    Do, Io = orc.search_preassigned(q, probes, keys, 10)
+    np.testing.assert_array_equal(I, Io)
+    np.testing.assert_array_equal(D, Do)  # BITWISE vs oracle
+
+
 def test_chunked_assign_matches_unchunked():
     # ws_mb=1 forces the multi-chunk assign/coarse paths (the round-1
     # negative-OOB regression lived there): results must be identical to
