@@ -166,6 +166,8 @@ def main():
                     help="-1: workload default; 0/1 override")
     ap.add_argument("--pq-lut-global", type=int, default=-1,
                     help="-1: auto (on at m>=32); 0/1 override")
+    ap.add_argument("--pq-lut-mb", type=int, default=96,
+                    help="LUT chunk budget (LLC residency)")
     args = ap.parse_args()
 
     import torch
@@ -207,7 +209,8 @@ def main():
             "pq_precomputed": (cfg.get("pq_precomputed", 0)
                                if args.pq_precomputed < 0
                                else args.pq_precomputed),
-            "pq_lut_global": args.pq_lut_global}
+            "pq_lut_global": args.pq_lut_global,
+            "pq_lut_mb": args.pq_lut_mb}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)

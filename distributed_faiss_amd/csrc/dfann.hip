@@ -197,6 +197,11 @@ struct dfann_index {
   int pq_lut_global = -1;  // spec "pq_lut_global": ADC LUTs built to HBM
                            // by k_pq_lut, scan stages them coalesced
                            // (-1 auto: on at m >= 32; 0 off; 1 force)
+  int pq_lut_mb = 96;  // spec "pq_lut_mb": LUT chunk budget. Small enough
+                       // to stay LLC-resident (256 MB MALL): the scan's
+                       // reads hit L2 and the next chunk overwrites the
+                       // same lines before they are ever written back,
+                       // so the LUT never round-trips HBM.
   int scan_fan = 1;     // spec "scan_fan": list-segment fan (experiment)  // spec "coarse_bf16": assign/coarse GEMMs on
                              // bf16 MFMA (~16x f32 rate) — approximate
                              // ranking path for huge nlist
@@ -487,6 +492,8 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->max_ppc = (int)json_int(js, "max_ppc", 256);
   h->pq_pre = json_int(js, "pq_precomputed", 0) != 0;
   h->pq_lut_global = (int)json_int(js, "pq_lut_global", -1);
+  h->pq_lut_mb = (int)json_int(js, "pq_lut_mb", 96);
+  if (h->pq_lut_mb < 1) h->pq_lut_mb = 1;
   h->scan_fan = (int)json_int(js, "scan_fan", 1);
   if (h->scan_fan < 1) h->scan_fan = 1;
   if (h->scan_fan > 16) h->scan_fan = 16;
@@ -821,10 +828,13 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   } else if (use_glut) {
     auto gk = rk ? (ip ? k_scan_pq_ip_g_rk : k_scan_pq_l2_g_rk)
                  : (ip ? k_scan_pq_ip_g : k_scan_pq_l2_g);
-    // chunk queries so the HBM LUT stays inside the workspace budget
+    // chunk queries so the LUT buffer stays LLC-resident (pq_lut_mb):
+    // the scan's LUT reads then hit L2, and the next chunk overwrites
+    // the same lines before writeback — no HBM round-trip for the LUTs
     size_t row_b = (size_t)nprobe * h->m * 1024;  // LUT bytes per query
+    int budget_mb = std::min(h->ws_mb, h->pq_lut_mb);
     int64_t qch =
-        std::max<int64_t>(1, (int64_t)(((size_t)h->ws_mb << 20) / row_b));
+        std::max<int64_t>(1, (int64_t)(((size_t)budget_mb << 20) / row_b));
     if (qch > nq) qch = nq;
     h->pq_lut_ws.ensure((size_t)qch * row_b);
     float *lutg = h->pq_lut_ws.as<float>();
