@@ -786,9 +786,11 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   // re-reads the full m*256*dsub codebook from L2 per (query, probe)
   // block and dominates the launch (measured 8.1 ms/launch at the
   // configs[3] shape, ~125 GB/s of algorithmic code bytes).
+  int glut_knob = h->pq_lut_global;
+  if (const char *e = getenv("DFANN_PQ_LUT_GLOBAL"))  // experiment override
+    glut_knob = atoi(e);
   bool use_glut = h->type == T_IVFPQ && !use_pre &&
-                  (h->pq_lut_global == 1 ||
-                   (h->pq_lut_global < 0 && h->m >= 32)) &&
+                  (glut_knob == 1 || (glut_knob < 0 && h->m >= 32)) &&
                   h->dsub <= 64;
   if (use_glut) fam_floats = h->m * 256;  // LUT only, no rbuf
   // segment fan (spec "scan_fan"): kept as an experiment knob — measured
@@ -832,7 +834,10 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
     // the scan's LUT reads then hit L2, and the next chunk overwrites
     // the same lines before writeback — no HBM round-trip for the LUTs
     size_t row_b = (size_t)nprobe * h->m * 1024;  // LUT bytes per query
-    int budget_mb = std::min(h->ws_mb, h->pq_lut_mb);
+    int lut_mb = h->pq_lut_mb;
+    if (const char *e = getenv("DFANN_PQ_LUT_MB"))  // experiment override
+      if (int v = atoi(e)) lut_mb = v;
+    int budget_mb = std::min(h->ws_mb, lut_mb);
     int64_t qch =
         std::max<int64_t>(1, (int64_t)(((size_t)budget_mb << 20) / row_b));
     if (qch > nq) qch = nq;
