@@ -42,6 +42,26 @@ def main():
     eng.nprobe = args.nprobe
     qb = q.contiguous()
     nq = qb.shape[0]
+    if os.environ.get("EXP_SKEW"):
+        import numpy as np
+        off, _, _ = eng.get_lists()
+        lens = np.diff(off)
+        probes, _ = eng.coarse(qb.cpu().numpy(), args.nprobe)
+        plens = lens[probes.reshape(-1)]
+        tot = plens.sum()
+        print(json.dumps({
+            "list_len": {"max": int(lens.max()), "mean": float(lens.mean()),
+                         "p99": int(np.percentile(lens, 99)),
+                         "p999": int(np.percentile(lens, 99.9)),
+                         "top10": np.sort(lens)[-10:].tolist()},
+            "probed": {"rows_per_step": int(tot),
+                       "max_block_rows": int(plens.max()),
+                       "frac_rows_in_gt2048":
+                           float(plens[plens > 2048].sum() / tot),
+                       "frac_rows_in_gt8192":
+                           float(plens[plens > 8192].sum() / tot),
+                       "p99_block_rows": int(np.percentile(plens, 99))},
+        }), file=sys.stderr, flush=True)
     for s in args.settings.split(";"):
         if s == "off":
             os.environ["DFANN_PQ_LUT_GLOBAL"] = "0"
