@@ -84,7 +84,7 @@ WORKLOADS = {
     "ivfpq_100m8_d768_m64": dict(
         type="ivfpq", d=768, n=12_500_000, nlist=65536, m=64, nbits=8,
         metric=1, nq=10_000, k=10, centers=12_500, sigma=0.5, latent=32,
-        coarse_bf16=1, max_ppc=64, ws_mb=2048,
+        coarse_bf16=1, max_ppc=64, ws_mb=2048, pq_lut_f16=1,
     ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
@@ -166,8 +166,10 @@ def main():
                     help="-1: workload default; 0/1 override")
     ap.add_argument("--pq-lut-global", type=int, default=-1,
                     help="-1: auto (on at m>=32); 0/1 override")
-    ap.add_argument("--pq-lut-mb", type=int, default=96,
-                    help="LUT chunk budget (LLC residency)")
+    ap.add_argument("--pq-lut-mb", type=int, default=2048,
+                    help="LUT chunk budget (bigger chunks win)")
+    ap.add_argument("--pq-lut-f16", type=int, default=-1,
+                    help="-1: workload default; 0/1 override")
     args = ap.parse_args()
 
     import torch
@@ -210,7 +212,9 @@ def main():
                                if args.pq_precomputed < 0
                                else args.pq_precomputed),
             "pq_lut_global": args.pq_lut_global,
-            "pq_lut_mb": args.pq_lut_mb}
+            "pq_lut_mb": args.pq_lut_mb,
+            "pq_lut_f16": (cfg.get("pq_lut_f16", 0)
+                           if args.pq_lut_f16 < 0 else args.pq_lut_f16)}
     eng = HipEngine(spec=spec)
     t0 = time.time()
     eng.train_dev(xb)

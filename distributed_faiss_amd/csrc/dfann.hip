@@ -197,11 +197,16 @@ struct dfann_index {
   int pq_lut_global = -1;  // spec "pq_lut_global": ADC LUTs built to HBM
                            // by k_pq_lut, scan stages them coalesced
                            // (-1 auto: on at m >= 32; 0 off; 1 force)
-  int pq_lut_mb = 96;  // spec "pq_lut_mb": LUT chunk budget. Small enough
-                       // to stay LLC-resident (256 MB MALL): the scan's
-                       // reads hit L2 and the next chunk overwrites the
-                       // same lines before they are ever written back,
-                       // so the LUT never round-trips HBM.
+  int pq_lut_mb = 2048;  // spec "pq_lut_mb": LUT chunk budget (measured:
+                         // bigger chunks win — launch concurrency beats
+                         // LLC residency; sweep in BASELINE.md ladder)
+  bool pq_lut_f16 = false;  // spec "pq_lut_f16": __half ADC tables —
+                            // halves the LUT HBM round-trip AND the scan
+                            // LDS (4 blocks/CU at m=64). APPROXIMATION
+                            // (faiss GpuIndexIVFPQ useFloat16LookupTables
+                            // equivalent): documented tolerance path,
+                            // off by default; the exact path stays the
+                            // parity contract.
   int scan_fan = 1;     // spec "scan_fan": list-segment fan (experiment)  // spec "coarse_bf16": assign/coarse GEMMs on
                              // bf16 MFMA (~16x f32 rate) — approximate
                              // ranking path for huge nlist
@@ -492,8 +497,9 @@ static dfann_index *create_from_spec(const std::string &js) {
   h->max_ppc = (int)json_int(js, "max_ppc", 256);
   h->pq_pre = json_int(js, "pq_precomputed", 0) != 0;
   h->pq_lut_global = (int)json_int(js, "pq_lut_global", -1);
-  h->pq_lut_mb = (int)json_int(js, "pq_lut_mb", 96);
+  h->pq_lut_mb = (int)json_int(js, "pq_lut_mb", 2048);
   if (h->pq_lut_mb < 1) h->pq_lut_mb = 1;
+  h->pq_lut_f16 = json_int(js, "pq_lut_f16", 0) != 0;
   h->scan_fan = (int)json_int(js, "scan_fan", 1);
   if (h->scan_fan < 1) h->scan_fan = 1;
   if (h->scan_fan > 16) h->scan_fan = 16;
@@ -789,10 +795,17 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   int glut_knob = h->pq_lut_global;
   if (const char *e = getenv("DFANN_PQ_LUT_GLOBAL"))  // experiment override
     glut_knob = atoi(e);
-  bool use_glut = h->type == T_IVFPQ && !use_pre &&
-                  (glut_knob == 1 || (glut_knob < 0 && h->m >= 32)) &&
-                  h->dsub <= 64;
-  if (use_glut) fam_floats = h->m * 256;  // LUT only, no rbuf
+  bool lut_f16 = h->pq_lut_f16;
+  if (const char *e = getenv("DFANN_PQ_LUT_F16"))  // experiment override
+    lut_f16 = atoi(e) != 0;
+  // f32 GLUT measured a wash vs the in-kernel build (BASELINE.md ladder),
+  // so auto engages only for the fp16-LUT approximation path where the
+  // halved round-trip wins; =1 still forces the f32 variant for A/B.
+  bool use_glut = h->type == T_IVFPQ && !use_pre && h->dsub <= 64 &&
+                  (glut_knob == 1 || (glut_knob != 0 && lut_f16));
+  lut_f16 = lut_f16 && use_glut;
+  if (use_glut)  // LUT only, no rbuf; fam_floats is in FLOAT units
+    fam_floats = lut_f16 ? h->m * 128 : h->m * 256;
   // segment fan (spec "scan_fan"): kept as an experiment knob — measured
   // NEGATIVE at the 10M SQ8 shape (per-block staging/extraction overhead
   // outweighs tail imbalance: 1936 -> 1339 GB/s at fan 8), so default 1.
@@ -828,12 +841,14 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        cand_d, cand_p, fam_floats, h->term2.as<float>(),
                        h->term3_ws.as<float>(), h->qn_ws.as<float>());
   } else if (use_glut) {
-    auto gk = rk ? (ip ? k_scan_pq_ip_g_rk : k_scan_pq_l2_g_rk)
-                 : (ip ? k_scan_pq_ip_g : k_scan_pq_l2_g);
-    // chunk queries so the LUT buffer stays LLC-resident (pq_lut_mb):
-    // the scan's LUT reads then hit L2, and the next chunk overwrites
-    // the same lines before writeback — no HBM round-trip for the LUTs
-    size_t row_b = (size_t)nprobe * h->m * 1024;  // LUT bytes per query
+    auto gk = lut_f16 ? (rk ? (ip ? k_scan_pq_ip_gh_rk : k_scan_pq_l2_gh_rk)
+                            : (ip ? k_scan_pq_ip_gh : k_scan_pq_l2_gh))
+                      : (rk ? (ip ? k_scan_pq_ip_g_rk : k_scan_pq_l2_g_rk)
+                            : (ip ? k_scan_pq_ip_g : k_scan_pq_l2_g));
+    // chunk queries so the LUT buffer stays inside the budget (measured:
+    // bigger chunks win — concurrency beats LLC residency)
+    size_t row_b =
+        (size_t)nprobe * h->m * 256 * (lut_f16 ? 2 : 4);  // bytes/query
     int lut_mb = h->pq_lut_mb;
     if (const char *e = getenv("DFANN_PQ_LUT_MB"))  // experiment override
       if (int v = atoi(e)) lut_mb = v;
@@ -850,11 +865,19 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
       long long qpn = (long long)nqc * nprobe;
       dim3 lg((unsigned)((qpn + PQ_LUT_QPT - 1) / PQ_LUT_QPT),
               (unsigned)h->m);
-      hipLaunchKernelGGL(k_pq_lut, lg, dim3(256), lut_lds, stream,
-                         q + q0 * h->d, h->centroids.as<float>(),
-                         h->codebooks.as<float>(), probes + q0 * nprobe,
-                         (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
-                         lutg);
+      if (lut_f16) {
+        hipLaunchKernelGGL(k_pq_lut_f16, lg, dim3(256), lut_lds, stream,
+                           q + q0 * h->d, h->centroids.as<float>(),
+                           h->codebooks.as<float>(), probes + q0 * nprobe,
+                           (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
+                           h->pq_lut_ws.as<__half>());
+      } else {
+        hipLaunchKernelGGL(k_pq_lut, lg, dim3(256), lut_lds, stream,
+                           q + q0 * h->d, h->centroids.as<float>(),
+                           h->codebooks.as<float>(), probes + q0 * nprobe,
+                           (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
+                           lutg);
+      }
       hipLaunchKernelGGL(gk, dim3((unsigned)(nqc * nprobe)), dim3(scan_bs),
                          lds, stream, q + q0 * h->d, h->centroids.as<float>(),
                          h->codebooks.as<float>(), h->sq_vmin.as<float>(),

@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
 #include <stdint.h>
+#include <type_traits>
 
 #define DFANN_BLOCK 256
 #define SEL_CAP 1024            // candidate buffer entries (k <= 512)
@@ -695,6 +696,8 @@ extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
 // accumulation ORDER is unchanged (sequential over the reduced axis,
 // contract off), so the oracle bit-exactness contract holds.
 
+// (float) cast: identity for f32 LUTs, h->f32 convert for the fp16-LUT
+// approximation path (pq_lut_f16) — the accumulation itself stays f32
 #define DFANN_PROC16_PQ(WV, G)                                                 \
   if ((G) < m) {                                                               \
     unsigned w0_ = (WV).x, w1_ = (WV).y, w2_ = (WV).z, w3_ = (WV).w;           \
@@ -702,7 +705,7 @@ extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
       if ((G) + b < m) {                                                       \
         unsigned word = (b < 4) ? w0_ : (b < 8) ? w1_ : (b < 12) ? w2_ : w3_;  \
         unsigned c = (word >> ((b & 3) * 8)) & 0xFFu;                          \
-        acc = acc + lut[((G) + b) * 256 + c];                                  \
+        acc = acc + (float)lut[((G) + b) * 256 + c];                           \
       }                                                                        \
     }                                                                          \
   }
@@ -752,14 +755,15 @@ extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
     }                                                                          \
   }
 
-template <int FAM, bool IS_IP>
+template <int FAM, bool IS_IP, bool L16 = false>
 __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
                                                const float *__restrict__ fam,
                                                int d, int m) {
   float acc = 0.f;
   const uint4 zero4 = {0, 0, 0, 0};
   if (FAM == 0) {
-    const float *lut = fam;
+    using LT = typename std::conditional<L16, __half, float>::type;
+    const LT *lut = reinterpret_cast<const LT *>(fam);
     for (int g0 = 0; g0 < m; g0 += 64) {
 #pragma clang fp contract(off)
       uint4 wa = *reinterpret_cast<const uint4 *>(cp + g0);
@@ -822,7 +826,7 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
 }
 
 template <int FAM, bool IS_IP, bool REGSEL, bool PRE = false,
-          bool GLUT = false>
+          bool GLUT = false, bool L16 = false>
 __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
@@ -883,10 +887,14 @@ __device__ void ivf_scan_body(
     // algorithmic code bytes at the configs[3] shape). Values are
     // BIT-IDENTICAL to the in-kernel path: k_pq_lut uses the same
     // sequential-t accumulation with contract off.
-    float4 *dst = reinterpret_cast<float4 *>(fam);
-    const float4 *src = reinterpret_cast<const float4 *>(
-        glut + ((size_t)bq * nprobe + bp) * ((size_t)m * 256));
-    int n4 = m * 64;
+    // L16: the LUT row is __half (pq_lut_f16 approximation — half the
+    // HBM round-trip and half the LDS, 4 blocks/CU at m=64)
+    size_t row_elems = (size_t)m * 256;
+    uint4 *dst = reinterpret_cast<uint4 *>(fam);
+    const uint4 *src = reinterpret_cast<const uint4 *>(
+        reinterpret_cast<const char *>(glut) +
+        ((size_t)bq * nprobe + bp) * row_elems * (L16 ? 2 : 4));
+    int n4 = (int)(row_elems * (L16 ? 2 : 4) / 16);
     for (int e = threadIdx.x; e < n4; e += blockDim.x) dst[e] = src[e];
   } else if (FAM == 0) {
     // rbuf (d floats) AFTER the LUT
@@ -1119,7 +1127,7 @@ __device__ void ivf_scan_body(
         long long pos = base + (long long)u * BS + threadIdx.x;
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
-          float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
+          float acc = scan_row_dist<FAM, IS_IP, L16>(cp, fam, d, m);
           float dist = IS_IP ? -(bias + acc) : (PRE ? bias + acc : acc);
           loc.push(dist, (unsigned)pos);
         }
@@ -1134,7 +1142,7 @@ __device__ void ivf_scan_body(
         long long pos = base + (long long)u * BS + threadIdx.x;
         if (pos < s1) {
           const uint8_t *cp = codes + pos * (size_t)stride;
-          float acc = scan_row_dist<FAM, IS_IP>(cp, fam, d, m);
+          float acc = scan_row_dist<FAM, IS_IP, L16>(cp, fam, d, m);
           float dist = IS_IP ? -(bias + acc) : (PRE ? bias + acc : acc);
           sel_try(s, dist, (unsigned)pos);
         }
@@ -1219,6 +1227,24 @@ INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g, true, false)
 INSTANTIATE_SCAN_GLUT(k_scan_pq_l2_g_rk, false, true)
 INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g_rk, true, true)
 
+// fp16-LUT variants (pq_lut_f16): glut holds __half rows
+#define INSTANTIATE_SCAN_GLUT_H(NAME, IS_IP, REGSEL)                           \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
+      unsigned *cand_p, int fam_floats, const float *glut) {                   \
+    ivf_scan_body<0, IS_IP, REGSEL, false, true, true>(                        \
+        q, cent, cb, sq_vmin, sq_scale, probes, keys, codes, off, nq, nprobe,  \
+        d, m, dsub, k, stride, cand_d, cand_p, fam_floats, nullptr, nullptr,   \
+        nullptr, 1, glut);                                                     \
+  }
+INSTANTIATE_SCAN_GLUT_H(k_scan_pq_l2_gh, false, false)
+INSTANTIATE_SCAN_GLUT_H(k_scan_pq_ip_gh, true, false)
+INSTANTIATE_SCAN_GLUT_H(k_scan_pq_l2_gh_rk, false, true)
+INSTANTIATE_SCAN_GLUT_H(k_scan_pq_ip_gh_rk, true, true)
+
 // ---------------------------------------------------------------------------
 // k_pq_lut: ADC lookup tables to HBM, one 256-entry row per (query,
 // probe, subspace) at out[(qp)*m*256 + j*256 + c] — the scan's GLUT
@@ -1232,10 +1258,11 @@ INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g_rk, true, true)
 // oracle parity preserved (oracle/core.py adc_scan).
 // ---------------------------------------------------------------------------
 #define PQ_LUT_QPT 64
-extern "C" __global__ __launch_bounds__(256) void k_pq_lut(
+template <typename LUTT>
+__device__ __forceinline__ void pq_lut_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const int *__restrict__ probes, int nq,
-    int nprobe, int d, int m, int dsub, int is_ip, float *__restrict__ out) {
+    int nprobe, int d, int m, int dsub, int is_ip, LUTT *__restrict__ out) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   int pad = dsub | 1;  // odd stride -> gcd(pad, banks) == 1
   float *cb_sm = reinterpret_cast<float *>(smem);  // 256 * pad
@@ -1276,8 +1303,22 @@ extern "C" __global__ __launch_bounds__(256) void k_pq_lut(
         acc = acc + diff * diff;
       }
     }
-    out[(size_t)(qp0 + rr) * ((size_t)m * 256) + (size_t)j * 256 + c] = acc;
+    out[(size_t)(qp0 + rr) * ((size_t)m * 256) + (size_t)j * 256 + c] =
+        (LUTT)acc;  // __half: round-nearest-even
   }
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_pq_lut(
+    const float *q, const float *cent, const float *cb, const int *probes,
+    int nq, int nprobe, int d, int m, int dsub, int is_ip, float *out) {
+  pq_lut_body<float>(q, cent, cb, probes, nq, nprobe, d, m, dsub, is_ip, out);
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_pq_lut_f16(
+    const float *q, const float *cent, const float *cb, const int *probes,
+    int nq, int nprobe, int d, int m, int dsub, int is_ip, __half *out) {
+  pq_lut_body<__half>(q, cent, cb, probes, nq, nprobe, d, m, dsub, is_ip,
+                      out);
 }
 INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
 INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)

@@ -63,10 +63,18 @@ def main():
                        "p99_block_rows": int(np.percentile(plens, 99))},
         }), file=sys.stderr, flush=True)
     for s in args.settings.split(";"):
+        # tokens: "off" | "<mb>" (f32 GLUT) | "f16" | "f16:<mb>"
         if s == "off":
             os.environ["DFANN_PQ_LUT_GLOBAL"] = "0"
-        else:
+            os.environ["DFANN_PQ_LUT_F16"] = "0"
+        elif s.startswith("f16"):
             os.environ["DFANN_PQ_LUT_GLOBAL"] = "-1"
+            os.environ["DFANN_PQ_LUT_F16"] = "1"
+            os.environ["DFANN_PQ_LUT_MB"] = (s.split(":", 1)[1]
+                                             if ":" in s else "2048")
+        else:
+            os.environ["DFANN_PQ_LUT_GLOBAL"] = "1"
+            os.environ["DFANN_PQ_LUT_F16"] = "0"
             os.environ["DFANN_PQ_LUT_MB"] = s
         for _ in range(args.warmup):
             eng.search_dev(qb, k)

@@ -391,6 +391,44 @@ def test_pq_glut_path_bitexact(m, metric):
     np.testing.assert_array_equal(D, Do)  # BITWISE vs oracle
 
 
+def test_pq_lut_f16_tolerance_path():
+    # fp16 ADC tables (pq_lut_f16 — the faiss GpuIndexIVFPQ
+    # useFloat16LookupTables equivalent): documented approximation, so
+    # tolerance + rank-overlap parity against the exact path on SHARED
+    # artifacts, plus bitwise self-determinism.
+    d, nlist, m = 64, 8, 32
+    cent, xb = _clustered(nlist, 300, d, seed=51)
+    rng = np.random.default_rng(52)
+    cb = rng.standard_normal((m, 256, d // m)).astype(np.float32)
+    lbl = rng.integers(0, nlist, xb.shape[0])
+    cw = rng.integers(0, 256, (xb.shape[0], m))
+    dec = np.concatenate([cb[j][cw[:, j]] for j in range(m)], axis=1)
+    xb = (cent[lbl] + dec
+          + 1e-3 * rng.standard_normal(xb.shape)).astype(np.float32)
+    q = (xb[::17][:30]
+         + 0.01 * rng.standard_normal((30, d))).astype(np.float32)
+    res = []
+    for f16 in (0, 1):
+        spec = {"type": "ivfpq", "dim": d, "metric": L2, "nlist": nlist,
+                "m": m, "nprobe": nlist, "seed": 7, "pq_lut_f16": f16}
+        eng = HipEngine(spec=spec)
+        eng.set_trained(cent, cb)
+        eng.add(xb)
+        probes, keys = eng.coarse(q, nlist)
+        res.append(eng.search_preassigned(q, probes, keys, 10))
+        if f16:  # determinism of the approximation path
+            D2, I2 = eng.search_preassigned(q, probes, keys, 10)
+            np.testing.assert_array_equal(res[-1][0], D2)
+            np.testing.assert_array_equal(res[-1][1], I2)
+    (D0, I0), (D1, I1) = res
+    overlap = np.mean([len(set(a) & set(b)) / len(a)
+                       for a, b in zip(I0, I1)])
+    assert overlap > 0.97, f"f16 vs exact id overlap {overlap}"
+    same = I0 == I1
+    assert same.mean() > 0.9
+    np.testing.assert_allclose(D0[same], D1[same], rtol=5e-3, atol=5e-3)
+
+
 def test_chunked_assign_matches_unchunked():
     # ws_mb=1 forces the multi-chunk assign/coarse paths (the round-1
     # negative-OOB regression lived there): results must be identical to
