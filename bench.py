@@ -170,6 +170,8 @@ def main():
                     help="LUT chunk budget (bigger chunks win)")
     ap.add_argument("--pq-lut-f16", type=int, default=-1,
                     help="-1: workload default; 0/1 override")
+    ap.add_argument("--graph", type=int, default=1,
+                    help="HIP-graph the serving step at N=1 (0: eager)")
     args = ap.parse_args()
 
     import torch
@@ -272,10 +274,32 @@ def main():
 
     for _ in range(args.warmup):
         run_search(nprobe, q)
+    # Serving-style step: capture the whole search step (coarse -> scan ->
+    # merge) in a HIP graph and replay it — identical kernels and work,
+    # zero host launch gaps (the 1M step is otherwise launch-bound).
+    # Verified below: one replay must reproduce the eager step bitwise.
+    # Multi-rank keeps the eager path (RCCL collectives stay outside
+    # graphs), so SCALE numbers are conservative.
+    use_graph = world == 1 and args.graph
+    if use_graph:
+        Dref, gref = run_search(nprobe, q)
+        gobj = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(gobj):
+            Dcap, gcap = run_search(nprobe, q)
+        gobj.replay()
+        torch.cuda.synchronize()
+        assert torch.equal(gcap, gref) and torch.equal(Dcap, Dref), \
+            "graph replay diverged from eager step"
+
+        def step():
+            gobj.replay()
+    else:
+        def step():
+            run_search(nprobe, q)
     barrier()
     t0 = time.time()
     for _ in range(args.steps):
-        run_search(nprobe, q)
+        step()
     barrier()
     elapsed = time.time() - t0
     if world > 1:
@@ -403,6 +427,11 @@ def main():
                 "batch": cfg["nq"],
                 "metric_space": "l2" if metric == 1 else "dot",
                 "coarse_dtype": "bf16" if cfg.get("coarse_bf16") else "f32",
+                "pq_lut": ("f16" if (cfg.get("pq_lut_f16", 0)
+                                     if args.pq_lut_f16 < 0
+                                     else args.pq_lut_f16)
+                           else "f32"),
+                "step_graph": bool(use_graph),
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
