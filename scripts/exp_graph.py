@@ -48,10 +48,25 @@ def main():
     try_capture("search_dev fresh out", lambda: eng.search_dev(q, 10))
     try_capture("coarse-only (dfann_coarse)", lambda: eng.search_preassigned.__self__ and None)
 
+    from distributed_faiss_amd.hip_engine import merge_topk_dev
+    D, I = eng.search_dev(q, 10)
+    Da, Ia = allgather_shard_topk(D, I)
+    torch.cuda.synchronize()
+    try_capture("merge_topk_dev only", lambda: merge_topk_dev(Da, Ia, 10, False))
+    Dm, slots = merge_topk_dev(Da, Ia, 10, False)
+    torch.cuda.synchronize()
+
+    def post():
+        s_idx = torch.div(slots, 2000 * 10, rounding_mode="floor")
+        local = Ia.reshape(-1)[slots]
+        return s_idx * 100_000 + local
+    try_capture("post indexing only", post)
+    try_capture("merge_gathered", lambda: merge_gathered(Da, Ia, 10, False))
+
     def full():
         D, I = eng.search_dev(q, 10)
-        Da, Ia = allgather_shard_topk(D, I)
-        merge_gathered(Da, Ia, 10, False)
+        Da2, Ia2 = allgather_shard_topk(D, I)
+        merge_gathered(Da2, Ia2, 10, False)
     try_capture("full step", full)
 
 
