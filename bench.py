@@ -235,16 +235,19 @@ def main():
     log(f"ground truth in {time.time()-t0:.1f}s")
 
     def run_search(nprobe, qt):
+        # serving step: results stay in HBM (no per-step D2H sync; the
+        # reference-API client path still returns host arrays)
         eng.nprobe = nprobe
         D, I = eng.search_dev(qt, k)
         Da, Ia = allgather_shard_topk(D, I)
-        Dm, s_i, loc = merge_gathered(Da, Ia, k, maximize)
+        Dm, s_i, loc = merge_gathered(Da, Ia, k, maximize, device_out=True)
         return Dm, s_i * cfg["n"] + loc
 
     def recall_at(nprobe, nq_eval=2048):
         qt = q[:nq_eval].contiguous()
         _, got = run_search(nprobe, qt)
         # faiss convention: true NN within returned top-k
+        got = got.cpu().numpy()
         hits = (got == gt_global[:nq_eval, :1]).any(axis=1)
         return float(hits.mean())
 

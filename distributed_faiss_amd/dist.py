@@ -54,11 +54,14 @@ def allgather_shard_topk(D_local, I_local):
     return torch.stack(Dg), torch.stack(Ig)
 
 
-def merge_gathered(Dall, Iall, k, maximize):
+def merge_gathered(Dall, Iall, k, maximize, device_out=False):
     """Merge (S,nq,k) shard results into (nq,k), reference heap semantics
     (returned distances NEGATED for maximize — quirk 2). Returns
     (D, shard_idx, local_ids): shard_idx/local_ids map each winner back to
-    its shard and per-shard id, mirroring client.py:290,297-298."""
+    its shard and per-shard id, mirroring client.py:290,297-298.
+    device_out=True keeps the three results in HBM (serving step: no
+    per-step D2H sync; graph-capturable) — the reference-API client path
+    always returns host arrays."""
     if Dall.is_cuda:
         import torch
 
@@ -70,6 +73,8 @@ def merge_gathered(Dall, Iall, k, maximize):
         # the GPU (host numpy here cost ~half the step time at 10k batch)
         s_idx = torch.div(slots, nq * kk, rounding_mode="floor")
         local = Iall.reshape(-1)[slots]
+        if device_out:
+            return Dm, s_idx, local
         return Dm.cpu().numpy(), s_idx.cpu().numpy(), local.cpu().numpy()
     # CPU (gloo tests): numpy restatement
     Da = Dall.numpy()
