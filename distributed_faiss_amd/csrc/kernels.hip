@@ -621,8 +621,21 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows_rk(
   RegTopK<16> loc;
   loc.init();
   const float *kp = keys + row * ldk;
-  for (long long c = threadIdx.x; c < cols; c += blockDim.x)
-    loc.push(kp[c], (unsigned)c + base);
+  // 8 independent loads in flight per iteration: the serial
+  // load->push->load chain left one HBM/L2 round trip per element
+  // (measured 1.6 ms for the 8192x65536 coarse top-k — latency, not
+  // bandwidth). push order per lane unchanged (c ascending).
+  const long long BS = blockDim.x;
+  long long c = threadIdx.x;
+  for (; c + 7 * BS < cols; c += 8 * BS) {
+    float v[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) v[u] = kp[c + u * BS];
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      loc.push(v[u], (unsigned)(c + u * BS) + base);
+  }
+  for (; c < cols; c += BS) loc.push(kp[c], (unsigned)c + base);
   __syncthreads();
   regtopk_block_extract<16>(loc, k, smem, out_d + row * ldo,
                             out_p + row * ldo);
@@ -895,7 +908,21 @@ __device__ void ivf_scan_body(
         reinterpret_cast<const char *>(glut) +
         ((size_t)bq * nprobe + bp) * row_elems * (L16 ? 2 : 4));
     int n4 = (int)(row_elems * (L16 ? 2 : 4) / 16);
-    for (int e = threadIdx.x; e < n4; e += blockDim.x) dst[e] = src[e];
+    // all loads issued before any LDS write: one HBM latency instead of
+    // a load->write->load chain (n4 is 2-8x blockDim at the shapes the
+    // GLUT gate admits; generic loop kept for odd shapes)
+    if (n4 % (int)blockDim.x == 0 && n4 / (int)blockDim.x <= 8) {
+      int rounds = n4 / (int)blockDim.x;
+      uint4 tmp[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        if (u < rounds) tmp[u] = src[u * blockDim.x + threadIdx.x];
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        if (u < rounds) dst[u * blockDim.x + threadIdx.x] = tmp[u];
+    } else {
+      for (int e = threadIdx.x; e < n4; e += blockDim.x) dst[e] = src[e];
+    }
   } else if (FAM == 0) {
     // rbuf (d floats) AFTER the LUT
     float *lut = fam;
@@ -1258,11 +1285,17 @@ INSTANTIATE_SCAN_GLUT_H(k_scan_pq_ip_gh_rk, true, true)
 // oracle parity preserved (oracle/core.py adc_scan).
 // ---------------------------------------------------------------------------
 #define PQ_LUT_QPT 64
-template <typename LUTT>
+// DSUB > 0: compile-time subspace width — the thread's codebook row is
+// held in REGISTERS across the row loop (the runtime-dsub build left it
+// in LDS and re-read it per row: 12-VGPR kernel, ~2.2 ms/step at the
+// configs[3] shape). DSUB == 0: runtime fallback.
+template <typename LUTT, int DSUB>
 __device__ __forceinline__ void pq_lut_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const int *__restrict__ probes, int nq,
-    int nprobe, int d, int m, int dsub, int is_ip, LUTT *__restrict__ out) {
+    int nprobe, int d, int m, int dsub_rt, int is_ip,
+    LUTT *__restrict__ out) {
+  const int dsub = DSUB > 0 ? DSUB : dsub_rt;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   int pad = dsub | 1;  // odd stride -> gcd(pad, banks) == 1
   float *cb_sm = reinterpret_cast<float *>(smem);  // 256 * pad
@@ -1291,34 +1324,76 @@ __device__ __forceinline__ void pq_lut_body(
   __syncthreads();
   int c = threadIdx.x;
   const float *crow = cb_sm + c * pad;
-  for (int rr = 0; rr < nrow; ++rr) {
-    const float *rs = r_sm + rr * pad;
-    float acc = 0.f;
-    for (int t = 0; t < dsub; ++t) {
+  if (DSUB > 0) {
+    // codebook row in registers (fully unrolled: no dynamic indexing);
+    // rs[t] reads are wave-broadcast (all lanes the same address).
+    // Same op order (sequential t, mul+add, contract off) — bit-exact.
+    float creg[DSUB > 0 ? DSUB : 1];
+#pragma unroll
+    for (int t = 0; t < DSUB; ++t) creg[t] = crow[t];
+    for (int rr = 0; rr < nrow; ++rr) {
+      const float *rs = r_sm + rr * pad;
+      float acc = 0.f;
+#pragma unroll
+      for (int t = 0; t < DSUB; ++t) {
 #pragma clang fp contract(off)
-      if (is_ip) {
-        acc = acc + rs[t] * crow[t];
-      } else {
-        float diff = rs[t] - crow[t];
-        acc = acc + diff * diff;
+        if (is_ip) {
+          acc = acc + rs[t] * creg[t];
+        } else {
+          float diff = rs[t] - creg[t];
+          acc = acc + diff * diff;
+        }
       }
+      out[(size_t)(qp0 + rr) * ((size_t)m * 256) + (size_t)j * 256 + c] =
+          (LUTT)acc;  // __half: round-nearest-even
     }
-    out[(size_t)(qp0 + rr) * ((size_t)m * 256) + (size_t)j * 256 + c] =
-        (LUTT)acc;  // __half: round-nearest-even
+  } else {
+    for (int rr = 0; rr < nrow; ++rr) {
+      const float *rs = r_sm + rr * pad;
+      float acc = 0.f;
+      for (int t = 0; t < dsub; ++t) {
+#pragma clang fp contract(off)
+        if (is_ip) {
+          acc = acc + rs[t] * crow[t];
+        } else {
+          float diff = rs[t] - crow[t];
+          acc = acc + diff * diff;
+        }
+      }
+      out[(size_t)(qp0 + rr) * ((size_t)m * 256) + (size_t)j * 256 + c] =
+          (LUTT)acc;  // __half: round-nearest-even
+    }
   }
 }
+
+#define DFANN_PQ_LUT_DISPATCH(LUTT, OUT)                                       \
+  switch (dsub) {                                                              \
+    case 2: pq_lut_body<LUTT, 2>(q, cent, cb, probes, nq, nprobe, d, m,        \
+                                 dsub, is_ip, OUT); break;                     \
+    case 4: pq_lut_body<LUTT, 4>(q, cent, cb, probes, nq, nprobe, d, m,        \
+                                 dsub, is_ip, OUT); break;                     \
+    case 6: pq_lut_body<LUTT, 6>(q, cent, cb, probes, nq, nprobe, d, m,        \
+                                 dsub, is_ip, OUT); break;                     \
+    case 8: pq_lut_body<LUTT, 8>(q, cent, cb, probes, nq, nprobe, d, m,        \
+                                 dsub, is_ip, OUT); break;                     \
+    case 12: pq_lut_body<LUTT, 12>(q, cent, cb, probes, nq, nprobe, d, m,      \
+                                   dsub, is_ip, OUT); break;                   \
+    case 16: pq_lut_body<LUTT, 16>(q, cent, cb, probes, nq, nprobe, d, m,      \
+                                   dsub, is_ip, OUT); break;                   \
+    default: pq_lut_body<LUTT, 0>(q, cent, cb, probes, nq, nprobe, d, m,       \
+                                  dsub, is_ip, OUT); break;                    \
+  }
 
 extern "C" __global__ __launch_bounds__(256) void k_pq_lut(
     const float *q, const float *cent, const float *cb, const int *probes,
     int nq, int nprobe, int d, int m, int dsub, int is_ip, float *out) {
-  pq_lut_body<float>(q, cent, cb, probes, nq, nprobe, d, m, dsub, is_ip, out);
+  DFANN_PQ_LUT_DISPATCH(float, out)
 }
 
 extern "C" __global__ __launch_bounds__(256) void k_pq_lut_f16(
     const float *q, const float *cent, const float *cb, const int *probes,
     int nq, int nprobe, int d, int m, int dsub, int is_ip, __half *out) {
-  pq_lut_body<__half>(q, cent, cb, probes, nq, nprobe, d, m, dsub, is_ip,
-                      out);
+  DFANN_PQ_LUT_DISPATCH(__half, out)
 }
 INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
 INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)
