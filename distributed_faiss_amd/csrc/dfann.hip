@@ -281,10 +281,18 @@ static void gemm_keys_b(dfann_index *h, const float *A, int64_t Mrows,
     h->ws_bf16a.ensure((size_t)Mrows * K * 2);
     hipLaunchKernelGGL(k_f32_to_bf16, grid1d(Mrows * K), dim3(256), 0, stream,
                        A, Mrows * K, h->ws_bf16a.as<unsigned short>());
-    auto bk = (K % 8 == 0) ? k_gemm_bf16_glds : k_gemm_bf16_nt;
-    hipLaunchKernelGGL(bk, g, dim3(256), 0, stream,
-                       h->ws_bf16a.as<unsigned short>(), B_bf16, keys,
-                       (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);
+    if (K % 8 == 0 && N >= 2048 && Mrows >= 256 && K >= 64) {
+      // big coarse/assign shapes: 256^2-tile glds kernel (512 threads)
+      dim3 g2((unsigned)((N + 255) / 256), (unsigned)((Mrows + 255) / 256));
+      hipLaunchKernelGGL(k_gemm_bf16_256, g2, dim3(512), 0, stream,
+                         h->ws_bf16a.as<unsigned short>(), B_bf16, keys,
+                         (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);
+    } else {
+      auto bk = (K % 8 == 0) ? k_gemm_bf16_glds : k_gemm_bf16_nt;
+      hipLaunchKernelGGL(bk, g, dim3(256), 0, stream,
+                         h->ws_bf16a.as<unsigned short>(), B_bf16, keys,
+                         (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);
+    }
   } else {
     hipLaunchKernelGGL(k_gemm_nt, g, dim3(256), 0, stream, A, B, keys,
                        (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);

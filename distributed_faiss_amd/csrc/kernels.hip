@@ -645,6 +645,104 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows_rk(
 // running argmin across key-matrix chunks (assignment). rows = points.
 // best_v/best_i persist across chunk calls (init by k_fill_assign_init).
 // Ties: lowest global column (chunks ascending, strict <).
+// 256^2-tile variant (cdna_hip_programming.md §5 glds table: 256² BK=64
+// 2-buffer glds is the top tier for large GEMMs — the 128² tile peaks
+// ~620 TF, this shape ~1.2 PF): 512 threads = 8 waves as 2(M)x4(N),
+// wave tile 128x64, acc 8x4 f32x4. Same source-XOR swizzle + lane-linear
+// LDS + single __shared__ as k_gemm_bf16_glds. Used for the big
+// coarse/assign GEMMs (N = nlist >= 2048); ragged M/N/K handled by the
+// stage guards (out-of-range positions stage zeros).
+#define GB2_T 256
+extern "C" __global__ __launch_bounds__(512) void k_gemm_bf16_256(
+    const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
+    const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
+  __shared__ unsigned short smem2[2][2][GB2_T][GB_K];
+  int bi = blockIdx.y * GB2_T;
+  int bj = blockIdx.x * GB2_T;
+  int tid = threadIdx.x;
+  int lane = tid & 63, w = tid >> 6;
+  int wr = (w >> 2) * 128, wc = (w & 3) * 64;
+  f32x4 acc[8][4] = {};
+  int li = lane & 15;
+  int ke = (lane >> 4) * 8;
+
+  // 2048 16-B groups per operand, 8 waves x 4 glds x 64 lanes
+#define GLDS2_STAGE(BUF, K0)                                                   \
+  _Pragma("unroll") for (int i = 0; i < 4; ++i) {                              \
+    int g = (w * 4 + i) * 64 + lane;                                           \
+    int r = g >> 3;                                                            \
+    int c8 = (g & 7) * 8;                                                      \
+    int c8s = c8 ^ ((r & 7) << 3);                                             \
+    unsigned short *ldst = &smem2[BUF][0][0][0] + ((size_t)(w * 4 + i) * 64) * 8; \
+    const unsigned short *ga = &A[(size_t)(bi + r) * lda + (K0) + c8s];        \
+    const unsigned short *gb = &B[(size_t)(bj + r) * ldb + (K0) + c8s];        \
+    bool oka = (bi + r < M) && ((K0) + c8s + 7 < K);                           \
+    bool okb = (bj + r < N) && ((K0) + c8s + 7 < K);                           \
+    if (oka)                                                                   \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) unsigned int *)ga,          \
+          (__attribute__((address_space(3))) unsigned int *)ldst, 16, 0, 0);   \
+    else                                                                       \
+      *reinterpret_cast<uint4 *>(                                              \
+          &smem2[BUF][0][0][0] + (size_t)g * 8) = uint4{0, 0, 0, 0};           \
+    unsigned short *ldstB = &smem2[BUF][1][0][0] + ((size_t)(w * 4 + i) * 64) * 8; \
+    if (okb)                                                                   \
+      __builtin_amdgcn_global_load_lds(                                        \
+          (const __attribute__((address_space(1))) unsigned int *)gb,          \
+          (__attribute__((address_space(3))) unsigned int *)ldstB, 16, 0, 0);  \
+    else                                                                       \
+      *reinterpret_cast<uint4 *>(                                              \
+          &smem2[BUF][1][0][0] + (size_t)g * 8) = uint4{0, 0, 0, 0};           \
+  }
+
+#define GLDS2_MFMA(BUF)                                                        \
+  _Pragma("unroll") for (int kk = 0; kk < GB_K; kk += 32) {                    \
+    _Pragma("unroll") for (int tj = 0; tj < 4; ++tj) {                         \
+      int rb = wc + tj * 16 + li;                                              \
+      int cbx = (kk + ke) ^ ((rb & 7) << 3);                                   \
+      bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(&smem2[BUF][1][rb][cbx]);  \
+      _Pragma("unroll") for (int ti = 0; ti < 8; ++ti) {                       \
+        int ra = wr + ti * 16 + li;                                            \
+        int ca = (kk + ke) ^ ((ra & 7) << 3);                                  \
+        bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&smem2[BUF][0][ra][ca]); \
+        acc[ti][tj] =                                                          \
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj],       \
+                                                    0, 0, 0);                  \
+      }                                                                        \
+    }                                                                          \
+  }
+
+  GLDS2_STAGE(0, 0)
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = GB_K; k0 < K; k0 += GB_K) {
+    GLDS2_STAGE(cur ^ 1, k0)
+    GLDS2_MFMA(cur)
+    __syncthreads();
+    cur ^= 1;
+  }
+  GLDS2_MFMA(cur)
+
+  int rrow = (lane >> 4) * 4;
+#pragma unroll
+  for (int ti = 0; ti < 8; ++ti) {
+#pragma unroll
+    for (int tj = 0; tj < 4; ++tj) {
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        int row = bi + wr + ti * 16 + rrow + rg;
+        int col = bj + wc + tj * 16 + li;
+        if (row < M && col < N)
+          C[(size_t)row * ldc + col] =
+              gemm_key(acc[ti][tj][rg], row, col, qn, bn, mode);
+      }
+    }
+  }
+#undef GLDS2_STAGE
+#undef GLDS2_MFMA
+}
+
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void k_assign_init(float *best_v, int *best_i, long long n) {
