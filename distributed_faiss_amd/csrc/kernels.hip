@@ -592,12 +592,32 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows(
   __syncthreads();
   const float *kp = keys + row * ldk;
   const int BS = blockDim.x;
-  for (long long c0 = 0; c0 < cols; c0 += (long long)2 * BS) {
-    sel_guard(s, k, 2 * BS);
+  if ((ldk & 3) == 0) {
+    // float4 row reads (see k_topk_rows_rk) — Sel is order-invariant
+    const float4 *kp4 = reinterpret_cast<const float4 *>(kp);
+    long long c4n = cols >> 2;
+    for (long long c0 = 0; c0 < c4n; c0 += BS) {
+      sel_guard(s, k, 4 * BS);
+      long long c4 = c0 + threadIdx.x;
+      if (c4 < c4n) {
+        float4 v4 = kp4[c4];
+        sel_try(s, v4.x, (unsigned)(c4 * 4 + 0) + base);
+        sel_try(s, v4.y, (unsigned)(c4 * 4 + 1) + base);
+        sel_try(s, v4.z, (unsigned)(c4 * 4 + 2) + base);
+        sel_try(s, v4.w, (unsigned)(c4 * 4 + 3) + base);
+      }
+    }
+    sel_guard(s, k, BS);
+    for (long long c = (c4n << 2) + threadIdx.x; c < cols; c += BS)
+      sel_try(s, kp[c], (unsigned)c + base);
+  } else {
+    for (long long c0 = 0; c0 < cols; c0 += (long long)2 * BS) {
+      sel_guard(s, k, 2 * BS);
 #pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      long long c = c0 + u * BS + threadIdx.x;
-      if (c < cols) sel_try(s, kp[c], (unsigned)c + base);
+      for (int u = 0; u < 2; ++u) {
+        long long c = c0 + u * BS + threadIdx.x;
+        if (c < cols) sel_try(s, kp[c], (unsigned)c + base);
+      }
     }
   }
   __syncthreads();
@@ -621,21 +641,39 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows_rk(
   RegTopK<16> loc;
   loc.init();
   const float *kp = keys + row * ldk;
-  // 8 independent loads in flight per iteration: the serial
-  // load->push->load chain left one HBM/L2 round trip per element
-  // (measured 1.6 ms for the 8192x65536 coarse top-k — latency, not
-  // bandwidth). push order per lane unchanged (c ascending).
+  // float4 row reads (1 KB per wave-load instead of 256 B) + 2 vectors
+  // in flight: the 4 B/lane version capped the 8192x65536 coarse top-k
+  // at ~1.5 TB/s. Lane->element assignment changes, result does not
+  // (RegTopK extraction orders globally by (dist, id)).
   const long long BS = blockDim.x;
-  long long c = threadIdx.x;
-  for (; c + 7 * BS < cols; c += 8 * BS) {
-    float v[8];
-#pragma unroll
-    for (int u = 0; u < 8; ++u) v[u] = kp[c + u * BS];
-#pragma unroll
-    for (int u = 0; u < 8; ++u)
-      loc.push(v[u], (unsigned)(c + u * BS) + base);
+  if ((ldk & 3) == 0) {  // float4-aligned rows
+    const float4 *kp4 = reinterpret_cast<const float4 *>(kp);
+    long long c4n = cols >> 2;
+    long long c4 = threadIdx.x;
+    for (; c4 + BS < c4n; c4 += 2 * BS) {
+      float4 v0 = kp4[c4], v1 = kp4[c4 + BS];
+      loc.push(v0.x, (unsigned)(c4 * 4 + 0) + base);
+      loc.push(v0.y, (unsigned)(c4 * 4 + 1) + base);
+      loc.push(v0.z, (unsigned)(c4 * 4 + 2) + base);
+      loc.push(v0.w, (unsigned)(c4 * 4 + 3) + base);
+      loc.push(v1.x, (unsigned)((c4 + BS) * 4 + 0) + base);
+      loc.push(v1.y, (unsigned)((c4 + BS) * 4 + 1) + base);
+      loc.push(v1.z, (unsigned)((c4 + BS) * 4 + 2) + base);
+      loc.push(v1.w, (unsigned)((c4 + BS) * 4 + 3) + base);
+    }
+    for (; c4 < c4n; c4 += BS) {
+      float4 v0 = kp4[c4];
+      loc.push(v0.x, (unsigned)(c4 * 4 + 0) + base);
+      loc.push(v0.y, (unsigned)(c4 * 4 + 1) + base);
+      loc.push(v0.z, (unsigned)(c4 * 4 + 2) + base);
+      loc.push(v0.w, (unsigned)(c4 * 4 + 3) + base);
+    }
+    for (long long c = (c4n << 2) + threadIdx.x; c < cols; c += BS)
+      loc.push(kp[c], (unsigned)c + base);
+  } else {
+    for (long long c = threadIdx.x; c < cols; c += BS)
+      loc.push(kp[c], (unsigned)c + base);
   }
-  for (; c < cols; c += BS) loc.push(kp[c], (unsigned)c + base);
   __syncthreads();
   regtopk_block_extract<16>(loc, k, smem, out_d + row * ldo,
                             out_p + row * ldo);
@@ -767,9 +805,30 @@ extern "C" __global__ __launch_bounds__(256) void k_assign_rowblock(
   const float *kp = keys + r * ldk;
   float bv = DFANN_FLT_MAX;
   int bi = 0x7FFFFFFF;
-  for (long long c = threadIdx.x; c < cols; c += blockDim.x) {
-    float v = kp[c];
-    if (v < bv || (v == bv && (int)c < bi)) { bv = v; bi = (int)c; }
+  // float4 row reads where aligned (1 KB/wave-load; see k_topk_rows_rk).
+  // (value, column) tie-break everywhere -> lane assignment irrelevant.
+  if ((ldk & 3) == 0) {
+    const float4 *kp4 = reinterpret_cast<const float4 *>(kp);
+    long long c4n = cols >> 2;
+    long long c4 = threadIdx.x;
+    for (; c4 < c4n; c4 += blockDim.x) {
+      float4 v4 = kp4[c4];
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        float v = t == 0 ? v4.x : t == 1 ? v4.y : t == 2 ? v4.z : v4.w;
+        int c = (int)(c4 * 4 + t);
+        if (v < bv || (v == bv && c < bi)) { bv = v; bi = c; }
+      }
+    }
+    for (long long c = (c4n << 2) + threadIdx.x; c < cols; c += blockDim.x) {
+      float v = kp[c];
+      if (v < bv || (v == bv && (int)c < bi)) { bv = v; bi = (int)c; }
+    }
+  } else {
+    for (long long c = threadIdx.x; c < cols; c += blockDim.x) {
+      float v = kp[c];
+      if (v < bv || (v == bv && (int)c < bi)) { bv = v; bi = (int)c; }
+    }
   }
 #pragma unroll
   for (int o = 32; o > 0; o >>= 1) {
