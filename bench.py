@@ -45,6 +45,7 @@ WORKLOADS = {
     "ivfpq_1m_d128_m16": dict(
         type="ivfpq", d=128, n=1_000_000, nlist=1024, m=16, nbits=8,
         metric=1, nq=10_000, k=10, centers=10_000, sigma=0.45, latent=12,
+        pq_lut_f16=1,
     ),
     # BASELINE.json configs[1] (ivf_simple 1M, dot) — parity/regression
     "ivfflat_1m_d128": dict(
