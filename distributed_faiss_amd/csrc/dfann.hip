@@ -818,7 +818,10 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   // NEGATIVE at the 10M SQ8 shape (per-block staging/extraction overhead
   // outweighs tail imbalance: 1936 -> 1339 GB/s at fan 8), so default 1.
   int fan = h->type != T_IVFPQ ? h->scan_fan : 1;
-  size_t lds = (size_t)fam_floats * 4 + (rk ? REGSEL_LDS_BYTES : SEL_LDS_BYTES);
+  // REGSEL extract scratch aliases the fam region (used only after the
+  // scan, behind a barrier) -> max, not sum (kernels.hip ivf_scan_body)
+  size_t lds = rk ? std::max((size_t)fam_floats * 4, (size_t)REGSEL_LDS_BYTES)
+                  : (size_t)fam_floats * 4 + SEL_LDS_BYTES;
   if (lds > 160 * 1024)
     throw std::runtime_error("scan LDS over budget (m too large)");
   // big-LDS blocks (m=64 LUTs: ~69 KB -> 2 blocks/CU) run 512 threads so

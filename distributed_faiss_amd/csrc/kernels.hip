@@ -1015,7 +1015,12 @@ __device__ void ivf_scan_body(
   // winners; the candidate merge sees (nprobe*fan*k) entries per query.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float *fam = reinterpret_cast<float *>(smem);
-  char *selbase = smem + (size_t)fam_floats * 4;
+  // REGSEL extraction scratch is used only AFTER the scan loop (behind
+  // a __syncthreads), so it ALIASES the fam region — the block's LDS is
+  // max(fam, scratch), not the sum (m=64 fp16 LUT: 33.4 -> 32 KB,
+  // 4 -> 5 blocks/CU; the scan measured occupancy-proportional). The
+  // LDS-buffer selection path uses its region DURING the scan: offset.
+  char *selbase = REGSEL ? smem : smem + (size_t)fam_floats * 4;
   long long blk = blockIdx.x;
   int bq = (int)(blk / ((long long)nprobe * fan));
   int rest = (int)(blk % ((long long)nprobe * fan));
