@@ -828,6 +828,13 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   // the CU still holds 4 waves/SIMD (register path only; the LDS-buffer
   // selection path is capacity-sized for 256)
   unsigned scan_bs = (rk && lds >= 32 * 1024) ? 512 : 256;
+  // 32-KB-LDS blocks (fp16 m=64 LUTs): 5 blocks/CU fit the LDS but
+  // 5x512 busts the 2048-thread cap — 384 threads (6 waves) admits all
+  // 5 (1920 threads, 30 waves): more independent per-block phases in
+  // flight on the same CU
+  if (rk && lds == 32 * 1024) scan_bs = 384;
+  if (const char *e = getenv("DFANN_SCAN_BS"))  // experiment override
+    if (int v = atoi(e)) scan_bs = (unsigned)((v / 64) * 64);
   h->ws3.ensure((size_t)nq * nprobe * fan * k * 4);
   h->ws4.ensure((size_t)nq * nprobe * fan * k * 4);
   cand_d = h->ws3.as<float>();
