@@ -225,21 +225,7 @@ struct dfann_index {
   DevBuf ws1, ws2, ws3, ws4, ws5, ws_bf16a, ws_bf16b;
   DevBuf cent_bf16;
   DevBuf term2, term3_ws, qn_ws;  // PQ-L2 precomputed tables
-  DevBuf pq_lut_ws;  // HBM ADC LUTs for the GLUT scan path (2 buffers)
-  // chunk pipeline: k_pq_lut(c+1) on the aux stream overlaps scan(c)
-  // on the caller's stream (write-heavy builder hides under the
-  // read-heavy scan); events fork/join for graph-capture legality
-  hipStream_t aux_stream = nullptr;
-  hipEvent_t pipe_fork = nullptr, pipe_lut[2] = {}, pipe_scan[2] = {};
-  void ensure_pipe() {
-    if (aux_stream) return;
-    HIP_CHECK(hipStreamCreateWithFlags(&aux_stream, hipStreamNonBlocking));
-    HIP_CHECK(hipEventCreateWithFlags(&pipe_fork, hipEventDisableTiming));
-    for (int i = 0; i < 2; ++i) {
-      HIP_CHECK(hipEventCreateWithFlags(&pipe_lut[i], hipEventDisableTiming));
-      HIP_CHECK(hipEventCreateWithFlags(&pipe_scan[i], hipEventDisableTiming));
-    }
-  }
+  DevBuf pq_lut_ws;  // HBM ADC LUTs for the GLUT scan path
 
   // timing
   bool timing = false;
@@ -252,14 +238,6 @@ struct dfann_index {
         (void)hipEventDestroy(e.a);
         (void)hipEventDestroy(e.b);
       }
-    if (aux_stream) {
-      (void)hipStreamDestroy(aux_stream);
-      (void)hipEventDestroy(pipe_fork);
-      for (int i = 0; i < 2; ++i) {
-        (void)hipEventDestroy(pipe_lut[i]);
-        (void)hipEventDestroy(pipe_scan[i]);
-      }
-    }
   }
 
   TimingEv ev_begin(hipStream_t s) {
@@ -894,44 +872,30 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
       if (int v = atoi(e)) lut_mb = v;
     int budget_mb = std::min(h->ws_mb, lut_mb);
     int64_t qch =
-        std::max<int64_t>(1, (int64_t)(((size_t)budget_mb << 20) / row_b / 2));
+        std::max<int64_t>(1, (int64_t)(((size_t)budget_mb << 20) / row_b));
     if (qch > nq) qch = nq;
-    // two LUT buffers: k_pq_lut(chunk c+1) runs on the aux stream while
-    // scan(chunk c) reads buffer c&1 on the caller's stream — the
-    // write-heavy builder hides under the read-heavy scan. Events fork
-    // from / join to the caller's stream (graph-capture legal).
-    h->ensure_pipe();
-    h->pq_lut_ws.ensure((size_t)2 * qch * row_b);
-    char *lut_base = h->pq_lut_ws.as<char>();
+    h->pq_lut_ws.ensure((size_t)qch * row_b);
+    float *lutg = h->pq_lut_ws.as<float>();
     int pad = h->dsub | 1;
     size_t lut_lds = (size_t)(256 + PQ_LUT_QPT) * pad * 4;
-    HIP_CHECK(hipEventRecord(h->pipe_fork, stream));
-    int c = 0;
-    for (int64_t q0 = 0; q0 < nq; q0 += qch, ++c) {
+    for (int64_t q0 = 0; q0 < nq; q0 += qch) {
       int64_t nqc = std::min<int64_t>(qch, nq - q0);
       long long qpn = (long long)nqc * nprobe;
       dim3 lg((unsigned)((qpn + PQ_LUT_QPT - 1) / PQ_LUT_QPT),
               (unsigned)h->m);
-      float *lutg = reinterpret_cast<float *>(lut_base + (size_t)(c & 1) * qch * row_b);
-      // buffer c&1 free once scan(c-2) is done; first two chunks fork
-      HIP_CHECK(hipStreamWaitEvent(
-          h->aux_stream, c < 2 ? h->pipe_fork : h->pipe_scan[c & 1], 0));
       if (lut_f16) {
-        hipLaunchKernelGGL(k_pq_lut_f16, lg, dim3(256), lut_lds,
-                           h->aux_stream, q + q0 * h->d,
-                           h->centroids.as<float>(),
+        hipLaunchKernelGGL(k_pq_lut_f16, lg, dim3(256), lut_lds, stream,
+                           q + q0 * h->d, h->centroids.as<float>(),
                            h->codebooks.as<float>(), probes + q0 * nprobe,
                            (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
-                           reinterpret_cast<__half *>(lutg));
+                           h->pq_lut_ws.as<__half>());
       } else {
-        hipLaunchKernelGGL(k_pq_lut, lg, dim3(256), lut_lds, h->aux_stream,
+        hipLaunchKernelGGL(k_pq_lut, lg, dim3(256), lut_lds, stream,
                            q + q0 * h->d, h->centroids.as<float>(),
                            h->codebooks.as<float>(), probes + q0 * nprobe,
                            (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
                            lutg);
       }
-      HIP_CHECK(hipEventRecord(h->pipe_lut[c & 1], h->aux_stream));
-      HIP_CHECK(hipStreamWaitEvent(stream, h->pipe_lut[c & 1], 0));
       hipLaunchKernelGGL(gk, dim3((unsigned)(nqc * nprobe)), dim3(scan_bs),
                          lds, stream, q + q0 * h->d, h->centroids.as<float>(),
                          h->codebooks.as<float>(), h->sq_vmin.as<float>(),
@@ -941,7 +905,6 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                          h->m, h->dsub, k, h->stride,
                          cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
                          fam_floats, lutg);
-      HIP_CHECK(hipEventRecord(h->pipe_scan[c & 1], stream));
     }
   } else {
     hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe * fan)),
