@@ -1451,7 +1451,12 @@ INSTANTIATE_SCAN_GLUT_H(k_scan_pq_ip_gh_rk, true, true)
 // held in REGISTERS across the row loop (the runtime-dsub build left it
 // in LDS and re-read it per row: 12-VGPR kernel, ~2.2 ms/step at the
 // configs[3] shape). DSUB == 0: runtime fallback.
-template <typename LUTT, int DSUB>
+// PAIRED (f16 only, DSUB>0): each thread computes TWO adjacent codes
+// and stores one __half2 — the 2-B per-lane stores of the plain f16
+// variant measured 1.7x slower per byte than 4-B stores (1.3 vs
+// 2.3 TB/s); half the wave covers row rr, the other half rr+1, so LDS
+// and occupancy are unchanged.
+template <typename LUTT, int DSUB, bool PAIRED = false>
 __device__ __forceinline__ void pq_lut_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const int *__restrict__ probes, int nq,
@@ -1484,6 +1489,42 @@ __device__ __forceinline__ void pq_lut_body(
     }
   }
   __syncthreads();
+  if (PAIRED && DSUB > 0) {
+    int half_id = threadIdx.x >> 7;       // which row of the rr pair
+    int c0 = (threadIdx.x & 127) * 2;     // two adjacent codes
+    float cr0[DSUB > 0 ? DSUB : 1], cr1[DSUB > 0 ? DSUB : 1];
+#pragma unroll
+    for (int t = 0; t < DSUB; ++t) {
+      cr0[t] = cb_sm[c0 * pad + t];
+      cr1[t] = cb_sm[(c0 + 1) * pad + t];
+    }
+    for (int rr2 = 0; rr2 < nrow; rr2 += 2) {
+      int rr = rr2 + half_id;
+      if (rr < nrow) {
+        const float *rs = r_sm + rr * pad;
+        float a0 = 0.f, a1 = 0.f;
+#pragma unroll
+        for (int t = 0; t < DSUB; ++t) {
+#pragma clang fp contract(off)
+          if (is_ip) {
+            a0 = a0 + rs[t] * cr0[t];
+            a1 = a1 + rs[t] * cr1[t];
+          } else {
+            float d0 = rs[t] - cr0[t];
+            float d1 = rs[t] - cr1[t];
+            a0 = a0 + d0 * d0;
+            a1 = a1 + d1 * d1;
+          }
+        }
+        __half2 hv;
+        hv.x = (__half)a0;  // round-nearest-even, same values as the
+        hv.y = (__half)a1;  // unpaired variant
+        reinterpret_cast<__half2 *>(out)[((size_t)(qp0 + rr) * m + j) * 128 +
+                                         (threadIdx.x & 127)] = hv;
+      }
+    }
+    return;
+  }
   int c = threadIdx.x;
   const float *crow = cb_sm + c * pad;
   if (DSUB > 0) {
@@ -1528,34 +1569,34 @@ __device__ __forceinline__ void pq_lut_body(
   }
 }
 
-#define DFANN_PQ_LUT_DISPATCH(LUTT, OUT)                                       \
+#define DFANN_PQ_LUT_DISPATCH(LUTT, PAIRED, OUT)                               \
   switch (dsub) {                                                              \
-    case 2: pq_lut_body<LUTT, 2>(q, cent, cb, probes, nq, nprobe, d, m,        \
-                                 dsub, is_ip, OUT); break;                     \
-    case 4: pq_lut_body<LUTT, 4>(q, cent, cb, probes, nq, nprobe, d, m,        \
-                                 dsub, is_ip, OUT); break;                     \
-    case 6: pq_lut_body<LUTT, 6>(q, cent, cb, probes, nq, nprobe, d, m,        \
-                                 dsub, is_ip, OUT); break;                     \
-    case 8: pq_lut_body<LUTT, 8>(q, cent, cb, probes, nq, nprobe, d, m,        \
-                                 dsub, is_ip, OUT); break;                     \
-    case 12: pq_lut_body<LUTT, 12>(q, cent, cb, probes, nq, nprobe, d, m,      \
-                                   dsub, is_ip, OUT); break;                   \
-    case 16: pq_lut_body<LUTT, 16>(q, cent, cb, probes, nq, nprobe, d, m,      \
-                                   dsub, is_ip, OUT); break;                   \
-    default: pq_lut_body<LUTT, 0>(q, cent, cb, probes, nq, nprobe, d, m,       \
-                                  dsub, is_ip, OUT); break;                    \
+    case 2: pq_lut_body<LUTT, 2, PAIRED>(q, cent, cb, probes, nq, nprobe, d,   \
+                                         m, dsub, is_ip, OUT); break;          \
+    case 4: pq_lut_body<LUTT, 4, PAIRED>(q, cent, cb, probes, nq, nprobe, d,   \
+                                         m, dsub, is_ip, OUT); break;          \
+    case 6: pq_lut_body<LUTT, 6, PAIRED>(q, cent, cb, probes, nq, nprobe, d,   \
+                                         m, dsub, is_ip, OUT); break;          \
+    case 8: pq_lut_body<LUTT, 8, PAIRED>(q, cent, cb, probes, nq, nprobe, d,   \
+                                         m, dsub, is_ip, OUT); break;          \
+    case 12: pq_lut_body<LUTT, 12, PAIRED>(q, cent, cb, probes, nq, nprobe,    \
+                                           d, m, dsub, is_ip, OUT); break;     \
+    case 16: pq_lut_body<LUTT, 16, PAIRED>(q, cent, cb, probes, nq, nprobe,    \
+                                           d, m, dsub, is_ip, OUT); break;     \
+    default: pq_lut_body<LUTT, 0, false>(q, cent, cb, probes, nq, nprobe, d,   \
+                                         m, dsub, is_ip, OUT); break;          \
   }
 
 extern "C" __global__ __launch_bounds__(256) void k_pq_lut(
     const float *q, const float *cent, const float *cb, const int *probes,
     int nq, int nprobe, int d, int m, int dsub, int is_ip, float *out) {
-  DFANN_PQ_LUT_DISPATCH(float, out)
+  DFANN_PQ_LUT_DISPATCH(float, false, out)
 }
 
 extern "C" __global__ __launch_bounds__(256) void k_pq_lut_f16(
     const float *q, const float *cent, const float *cb, const int *probes,
     int nq, int nprobe, int d, int m, int dsub, int is_ip, __half *out) {
-  DFANN_PQ_LUT_DISPATCH(__half, out)
+  DFANN_PQ_LUT_DISPATCH(__half, true, out)
 }
 INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
 INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)
