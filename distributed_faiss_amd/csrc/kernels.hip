@@ -1070,20 +1070,15 @@ __device__ void ivf_scan_body(
         reinterpret_cast<const char *>(glut) +
         ((size_t)bq * nprobe + bp) * row_elems * (L16 ? 2 : 4));
     int n4 = (int)(row_elems * (L16 ? 2 : 4) / 16);
-    // all loads issued before any LDS write: one HBM latency instead of
-    // a load->write->load chain (n4 is 2-8x blockDim at the shapes the
-    // GLUT gate admits; generic loop kept for odd shapes)
-    if (n4 % (int)blockDim.x == 0 && n4 / (int)blockDim.x <= 8) {
-      int rounds = n4 / (int)blockDim.x;
-      uint4 tmp[8];
-#pragma unroll
-      for (int u = 0; u < 8; ++u)
-        if (u < rounds) tmp[u] = src[u * blockDim.x + threadIdx.x];
-#pragma unroll
-      for (int u = 0; u < 8; ++u)
-        if (u < rounds) dst[u * blockDim.x + threadIdx.x] = tmp[u];
-    } else {
-      for (int e = threadIdx.x; e < n4; e += blockDim.x) dst[e] = src[e];
+    // glds: 16-B global->LDS DMA, no VGPR round-trip — the copy is
+    // lane-linear (lane i of each pass writes base + i*16), exactly the
+    // wave-uniform-base + lane x size form the DMA requires; the
+    // staging barrier below drains it (vmcnt(0) under outstanding glds)
+    for (int e = threadIdx.x; e < n4; e += blockDim.x) {
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(src + e),
+          (__attribute__((address_space(3))) unsigned int *)(dst + e), 16, 0,
+          0);
     }
   } else if (FAM == 0) {
     // rbuf (d floats) AFTER the LUT
