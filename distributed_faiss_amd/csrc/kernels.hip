@@ -1485,37 +1485,46 @@ __device__ __forceinline__ void pq_lut_body(
   }
   __syncthreads();
   if (PAIRED && DSUB > 0) {
-    int half_id = threadIdx.x >> 7;       // which row of the rr pair
-    int c0 = (threadIdx.x & 127) * 2;     // two adjacent codes
-    float cr0[DSUB > 0 ? DSUB : 1], cr1[DSUB > 0 ? DSUB : 1];
+    // four adjacent codes per thread = four INDEPENDENT accumulation
+    // chains (the 2-chain variant measured 63% issue-stall — the
+    // contract-off sequential adds are a 4-cycle VALU dependency chain)
+    // and one 8-B store; wave w covers row rr4+w. Per-entry op order
+    // unchanged (sequential t) — values identical to the scalar build.
+    int w4 = threadIdx.x >> 6;            // wave id: row within rr group
+    int c0 = (threadIdx.x & 63) * 4;      // four adjacent codes
+    float cr[4][DSUB > 0 ? DSUB : 1];
 #pragma unroll
-    for (int t = 0; t < DSUB; ++t) {
-      cr0[t] = cb_sm[c0 * pad + t];
-      cr1[t] = cb_sm[(c0 + 1) * pad + t];
-    }
-    for (int rr2 = 0; rr2 < nrow; rr2 += 2) {
-      int rr = rr2 + half_id;
+    for (int cc = 0; cc < 4; ++cc)
+#pragma unroll
+      for (int t = 0; t < DSUB; ++t) cr[cc][t] = cb_sm[(c0 + cc) * pad + t];
+    for (int rr4 = 0; rr4 < nrow; rr4 += 4) {
+      int rr = rr4 + w4;
       if (rr < nrow) {
         const float *rs = r_sm + rr * pad;
-        float a0 = 0.f, a1 = 0.f;
+        float a[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int t = 0; t < DSUB; ++t) {
 #pragma clang fp contract(off)
-          if (is_ip) {
-            a0 = a0 + rs[t] * cr0[t];
-            a1 = a1 + rs[t] * cr1[t];
-          } else {
-            float d0 = rs[t] - cr0[t];
-            float d1 = rs[t] - cr1[t];
-            a0 = a0 + d0 * d0;
-            a1 = a1 + d1 * d1;
+#pragma unroll
+          for (int cc = 0; cc < 4; ++cc) {
+            if (is_ip) {
+              a[cc] = a[cc] + rs[t] * cr[cc][t];
+            } else {
+              float dd = rs[t] - cr[cc][t];
+              a[cc] = a[cc] + dd * dd;
+            }
           }
         }
-        __half2 hv;
-        hv.x = (__half)a0;  // round-nearest-even, same values as the
-        hv.y = (__half)a1;  // unpaired variant
-        reinterpret_cast<__half2 *>(out)[((size_t)(qp0 + rr) * m + j) * 128 +
-                                         (threadIdx.x & 127)] = hv;
+        __half2 lo, hi;
+        lo.x = (__half)a[0];  // round-nearest-even, same values as the
+        lo.y = (__half)a[1];  // unpaired variant
+        hi.x = (__half)a[2];
+        hi.y = (__half)a[3];
+        uint2 pk;
+        pk.x = *reinterpret_cast<unsigned *>(&lo);
+        pk.y = *reinterpret_cast<unsigned *>(&hi);
+        reinterpret_cast<uint2 *>(out)[((size_t)(qp0 + rr) * m + j) * 64 +
+                                       (threadIdx.x & 63)] = pk;
       }
     }
     return;
