@@ -827,12 +827,16 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   // big-LDS blocks (m=64 LUTs: ~69 KB -> 2 blocks/CU) run 512 threads so
   // the CU still holds 4 waves/SIMD (register path only; the LDS-buffer
   // selection path is capacity-sized for 256)
-  unsigned scan_bs = (rk && lds >= 32 * 1024) ? 512 : 256;
-  // 32-KB-LDS blocks (fp16 m=64 LUTs): 5 blocks/CU fit the LDS but
-  // 5x512 busts the 2048-thread cap — 384 threads (6 waves) admits all
-  // 5 (1920 threads, 30 waves): more independent per-block phases in
-  // flight on the same CU
-  if (rk && lds == 32 * 1024) scan_bs = 384;
+  unsigned scan_bs = (rk && lds > 32 * 1024) ? 512 : 256;
+  // GLUT scans: SMALL blocks win (measured sweep, BASELINE.md ladder) —
+  // more independent blocks per CU means the per-block phases (LUT
+  // staging burst -> rows -> extract) de-correlate and the memory
+  // system stays busy: 1M f16 (8 KB LDS) 128 threads = 16 blocks/CU
+  // (+27% whole-step); configs[3] f16 (32 KB) 256 threads = 5 blocks
+  // (+8% over 384). The wave-parked fraction was 53% (profiles/r01
+  // SQ wait decomposition) — this is the phase-diversity lever.
+  if (use_glut && rk)
+    scan_bs = lds <= 16 * 1024 ? 128 : (lds <= 32 * 1024 ? 256 : 512);
   if (const char *e = getenv("DFANN_SCAN_BS"))  // experiment override
     if (int v = atoi(e)) scan_bs = (unsigned)((v / 64) * 64);
   h->ws3.ensure((size_t)nq * nprobe * fan * k * 4);
