@@ -63,7 +63,10 @@ def main():
                        "p99_block_rows": int(np.percentile(plens, 99))},
         }), file=sys.stderr, flush=True)
     for s in args.settings.split(";"):
-        # tokens: "off" | "<mb>" (f32 GLUT) | "f16" | "f16:<mb>"
+        # tokens: "off" | "<mb>" (f32 GLUT) | "f16" | "f16:<mb>" | "bs:<n>"
+        if s.startswith("bs:"):
+            os.environ["DFANN_SCAN_BS"] = s.split(":", 1)[1]
+            s = "off"
         if s == "off":
             os.environ["DFANN_PQ_LUT_GLOBAL"] = "0"
             os.environ["DFANN_PQ_LUT_F16"] = "0"
