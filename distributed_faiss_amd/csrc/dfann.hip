@@ -827,16 +827,16 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   // big-LDS blocks (m=64 LUTs: ~69 KB -> 2 blocks/CU) run 512 threads so
   // the CU still holds 4 waves/SIMD (register path only; the LDS-buffer
   // selection path is capacity-sized for 256)
-  unsigned scan_bs = (rk && lds > 32 * 1024) ? 512 : 256;
-  // GLUT scans: SMALL blocks win (measured sweep, BASELINE.md ladder) —
-  // more independent blocks per CU means the per-block phases (LUT
-  // staging burst -> rows -> extract) de-correlate and the memory
-  // system stays busy: 1M f16 (8 KB LDS) 128 threads = 16 blocks/CU
-  // (+27% whole-step); configs[3] f16 (32 KB) 256 threads = 5 blocks
-  // (+8% over 384). The wave-parked fraction was 53% (profiles/r01
-  // SQ wait decomposition) — this is the phase-diversity lever.
-  if (use_glut && rk)
-    scan_bs = lds <= 16 * 1024 ? 128 : (lds <= 32 * 1024 ? 256 : 512);
+  // Register-topk scans: SMALL blocks win (measured sweeps, BASELINE.md
+  // ladder) — more independent blocks per CU de-correlates the
+  // per-block phases (staging burst -> rows -> extract) and keeps the
+  // memory system busy (the SQ wait decomposition showed 53% of scan
+  // wave cycles parked): 1M f16 GLUT (8 KB LDS) 128 threads =
+  // 16 blocks/CU (+27% step); configs[3] f16 (32 KB) 256 = 5 blocks
+  // (+8% over 384); SQ8 10M 128 threads +8%. The LDS-buffer selection
+  // path keeps 256 (Sel capacity is sized for it).
+  unsigned scan_bs = 256;
+  if (rk) scan_bs = lds <= 16 * 1024 ? 128 : (lds <= 32 * 1024 ? 256 : 512);
   if (const char *e = getenv("DFANN_SCAN_BS"))  // experiment override
     if (int v = atoi(e)) scan_bs = (unsigned)((v / 64) * 64);
   h->ws3.ensure((size_t)nq * nprobe * fan * k * 4);
