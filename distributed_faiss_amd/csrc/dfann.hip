@@ -737,9 +737,9 @@ static void coarse_impl(dfann_index *h, int64_t nq, const float *q, int nprobe,
                 h->cnorm.as<float>(), nullptr, h->metric == M_IP ? 0 : 1, sc,
                 stream);
     if (use_regsel(nprobe)) {
-      // 128 threads: 16 blocks/CU (the phase-diversity lever, same as
-      // the scans); the extract handles any wave count
-      unsigned tk_bs = 128;
+      // 256 measured best (128 is -3%: this kernel is one streaming
+      // phase — no phase diversity to recover, unlike the scans)
+      unsigned tk_bs = 256;
       if (const char *e = getenv("DFANN_TOPK_BS"))  // experiment override
         if (int v = atoi(e)) tk_bs = (unsigned)((v / 64) * 64);
       hipLaunchKernelGGL(k_topk_rows_rk, dim3((unsigned)c), dim3(tk_bs),
