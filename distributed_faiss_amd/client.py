@@ -220,9 +220,15 @@ class IndexClient:
         filter_pos: int = -1,
         filter_value=None,
     ) -> Tuple[np.ndarray, List[List[object]]]:
-        # reference client.py:213-263: over-fetch x3, host-side post-filter
+        # reference client.py:213-263: over-fetch x3, host-side post-filter.
+        # The over-fetch is clamped to the engine's k cap (512, see
+        # include/dfann.h Limits) so a filtered search with top_k >= 171
+        # degrades to a smaller candidate pool instead of raising where the
+        # faiss-backed reference would succeed.
         filter_top_factor = 3
         actual_top_k = filter_top_factor * top_k if filter_pos >= 0 else top_k
+        if filter_pos >= 0 and actual_top_k > 512:
+            actual_top_k = max(top_k, 512)
         (scores, meta) = self.search(query, actual_top_k, index_id)
         if filter_pos < 0:
             return scores, meta

@@ -1020,7 +1020,18 @@ static void search_impl(dfann_index *h, int64_t nq, const float *q, int k,
   if (h->ntotal == 0) { pad_fill(h, nq, k, D, I, stream); return; }
   finalize_csr(h, stream);
   int nprobe = std::min(h->nprobe, h->nlist);
-  if (nprobe > 512) nprobe = 512;
+  if (nprobe > 512) {
+    // documented clamp (include/dfann.h Limits): warn once per process
+    static bool warned = false;
+    if (!warned) {
+      warned = true;
+      fprintf(stderr,
+              "[dfann] warning: nprobe=%d exceeds the engine cap; clamped to "
+              "512 (the faiss-backed reference would probe more lists)\n",
+              nprobe);
+    }
+    nprobe = 512;
+  }
   // probes + keys
   DevBuf &pb = h->ws2;
   pb.ensure((size_t)nq * nprobe * 8);

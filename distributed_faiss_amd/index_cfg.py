@@ -66,6 +66,16 @@ class IndexCfg:
     def from_json(cls, json_path):
         with open(json_path, "r") as f:
             kwargs = json.load(f)
+        # to_json_string serializes `extra` as a top-level key (it dumps
+        # __dict__); feeding it back through **kwargs would NEST it
+        # (cfg.extra == {"extra": {...}}), silently dropping engine knobs
+        # like code_size/bits_per_vector/seed on a train-from-round-trip.
+        # Flatten it back into kwargs — the on-disk format and the
+        # reference surface are unchanged.
+        extra = kwargs.pop("extra", None)
+        if isinstance(extra, dict):
+            for k, v in extra.items():
+                kwargs.setdefault(k, v)
         return cls(**kwargs)
 
     def to_json_string(self):

@@ -17,6 +17,18 @@
  *    I = -1 with D = +FLT_MAX (L2) / -FLT_MAX (IP).
  *  - ids are implicit arrival positions per index (reference quirk,
  *    SURVEY.md §2 item 9).
+ *
+ * Limits (engine caps; the faiss-backed reference has none of these):
+ *  - k (top-k) <= 512 everywhere (dfann_search*, dfann_merge_topk):
+ *    selection buffers are sized SEL_CAP=1024 with k<=512 headroom.
+ *    Larger k returns an error ("k > 512 unsupported").
+ *  - nprobe is clamped to min(nlist, 512). A request above 512 is
+ *    honored as 512 and a one-time warning is printed to stderr (the
+ *    reference would probe more lists; recall at nprobe=512 is the cap).
+ *  - per-shard ntotal < 2^32: ids round-trip through u32 inside the
+ *    scan/merge kernels (CSR positions and merge slots are u32), so one
+ *    shard holds at most 4.29e9 vectors. The C-ABI keeps i64 ids; the
+ *    1B-vector headline config is 8 shards x 125M, well inside the cap.
  */
 #ifndef DFANN_H
 #define DFANN_H
@@ -57,7 +69,7 @@ int dfann_add(dfann_index *h, int64_t n, const float *x_dev,
 /* --- search path ------------------------------------------------------ */
 
 /* Replaces faiss `Index.search` at ref index.py:257. D_dev: (nq,k) f32,
- * I_dev: (nq,k) i64. */
+ * I_dev: (nq,k) i64. k <= 512 (see Limits above). */
 int dfann_search(dfann_index *h, int64_t nq, const float *q_dev, int k,
                  float *D_dev, int64_t *I_dev, dfann_stream stream);
 
@@ -83,7 +95,9 @@ int dfann_search_preassigned(dfann_index *h, int64_t nq, const float *q_dev,
 
 /* --- knobs / introspection -------------------------------------------- */
 
-int dfann_set_nprobe(dfann_index *h, int nprobe); /* ref index.py:352-356 */
+/* ref index.py:352-356. Effective nprobe is min(nprobe, nlist, 512) at
+ * search time (see Limits above); values above 512 warn once. */
+int dfann_set_nprobe(dfann_index *h, int nprobe);
 int64_t dfann_ntotal(dfann_index *h);             /* faiss .ntotal */
 int dfann_nlist(dfann_index *h);                  /* faiss .nlist */
 int dfann_is_trained(dfann_index *h);
@@ -124,7 +138,8 @@ int dfann_get_sq_params(dfann_index *h, float *vmin_host,
  * I_dev: (S,nq,k) i64 shard ids (-1 pads). maximize: 1 for dot.
  * Output ids are GLOBAL slots s*nq*k + q*k + j into the gathered input
  * (the caller maps slots to (shard, local id) / metadata, mirroring ref
- * client.py:290,297-298). Dout is negated for maximize (ref quirk 2). */
+ * client.py:290,297-298). Dout is negated for maximize (ref quirk 2).
+ * k <= 512 (see Limits above). */
 int dfann_merge_topk(int64_t nq, int S, int k, const float *D_dev,
                      const int64_t *I_dev, int maximize, float *Dout_dev,
                      int64_t *Iout_dev, dfann_stream stream);

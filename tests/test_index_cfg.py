@@ -45,6 +45,24 @@ def test_json_round_trip(tmp_path):
     assert cfg2.nprobe == 4 and cfg2.train_num == 100
 
 
+def test_from_json_flattens_extra(tmp_path):
+    # to_json_string serializes cfg.extra as a top-level "extra" key;
+    # from_json must flatten it back so engine knobs (code_size,
+    # bits_per_vector, seed — read by engine_spec at train time) survive
+    # a cfg.json round trip instead of nesting as extra["extra"].
+    cfg = IndexCfg(index_builder_type="knnlm", dim=128, centroids=64,
+                   metric="l2", code_size=16, bits_per_vector=8, seed=7)
+    p = tmp_path / "cfg.json"
+    p.write_text(cfg.to_json_string())
+    cfg2 = IndexCfg.from_json(str(p))
+    assert cfg2.extra["code_size"] == 16
+    assert cfg2.extra["bits_per_vector"] == 8
+    assert cfg2.extra["seed"] == 7
+    assert "extra" not in cfg2.extra
+    spec = resolve_engine_spec(cfg2, 1000)
+    assert spec["m"] == 16 and spec["seed"] == 7
+
+
 def test_metric_mapping():
     assert IndexCfg(metric="dot").get_metric() == METRIC_INNER_PRODUCT
     assert IndexCfg(metric="l2").get_metric() == METRIC_L2
