@@ -71,13 +71,47 @@ class ResultHeap:
             self.I[i] = allI[i, order]
 
 
-class IndexClient:
-    """Manages a set of shard sub-indexes; searches merge client-side."""
+def _dist_ctx():
+    """(rank, world) of the active torch.distributed group, (0, 1) when
+    none. Import is lazy — the in-process client path works without
+    torch."""
+    try:
+        import torch.distributed as td
 
-    def __init__(self, server_list_path=None, cfg_path: Optional[str] = None, servers=None):
+        if td.is_available() and td.is_initialized():
+            return td.get_rank(), td.get_world_size()
+    except ImportError:
+        pass
+    return 0, 1
+
+
+class IndexClient:
+    """Manages a set of shard sub-indexes; searches merge client-side.
+
+    Two topologies behind the SAME surface:
+      * in-process (reference-shaped): N IndexServer objects, thread-pool
+        fan-out, numpy ResultHeap merge.
+      * distributed (MI355X-native, BASELINE.json north_star): one rank
+        per GPU shard under torch.distributed, auto-detected when the
+        process group is initialized with world > 1 and the client was
+        handed exactly its LOCAL server. search() then runs: local shard
+        search -> all-gather of per-shard (distance, id) top-k (RCCL
+        over xGMI on GPU, gloo on CPU) -> k-way merge (on-GPU
+        dfann_merge_topk / numpy) -> metadata mapped from (shard, slot)
+        exactly as the reference maps synthetic ids
+        (ref client.py:290,297-298). Merge semantics (incl. quirk-2 dot
+        negation) are identical to the in-process path, and search_dev()
+        exposes the same pipeline with device-resident results — the
+        bench's timed step IS this client path.
+    """
+
+    def __init__(self, server_list_path=None, cfg_path: Optional[str] = None,
+                 servers=None, distributed: Optional[bool] = None):
         """Either pass `servers` (list of in-process IndexServer) or
         `server_list_path` (reference format: first line = count, then
-        host,port lines — resolved against the in-process port registry)."""
+        host,port lines — resolved against the in-process port registry).
+        `distributed` overrides the auto-detection described above."""
+        self._dist_rank, self._dist_world = _dist_ctx()
         if servers is not None:
             self.sub_indexes: List[IndexServer] = list(servers)
         else:
@@ -92,6 +126,11 @@ class IndexClient:
 
                     self.sub_indexes.append(TcpClient(host, port))
         self.num_indexes = len(self.sub_indexes)
+        if distributed is None:
+            distributed = self._dist_world > 1 and self.num_indexes == 1
+        self.dist_mode = bool(distributed)
+        # logical shard count: ranks in dist mode, local servers otherwise
+        self.num_shards = self._dist_world if self.dist_mode else self.num_indexes
 
         index_ranks = [idx.get_rank() for idx in self.sub_indexes]
         self.index_rank_to_id = {
@@ -164,6 +203,10 @@ class IndexClient:
         if force_reload:
             self.pool.map(lambda idx: idx.drop_index(index_id), self.sub_indexes)
         all_loaded = self.pool.map(lambda idx: idx.load_index(index_id, cfg), self.sub_indexes)
+        if self.dist_mode:
+            from .dist import all_gather_object
+
+            all_loaded = all_gather_object(all_loaded[0])
         self.cfg = setup_cfg(cfg)
         if all(all_loaded):
             return True
@@ -187,6 +230,19 @@ class IndexClient:
         train_async_if_triggered: bool = True,
     ) -> None:
         # round-robin placement, random start (reference client.py:174-192)
+        if self.dist_mode:
+            # SPMD: every rank sees the same batch stream; batch i lands
+            # on shard i % world. Deterministic start 0 (deviation from
+            # the reference's random start — ranks must agree).
+            if index_id not in self.cur_server_ids:
+                self.cur_server_ids[index_id] = 0
+            if self.cur_server_ids[index_id] == self._dist_rank:
+                self.sub_indexes[0].add_index_data(
+                    index_id, embeddings, metadata, train_async_if_triggered
+                )
+            self.cur_server_ids[index_id] = (
+                self.cur_server_ids[index_id] + 1) % self.num_shards
+            return
         if index_id not in self.cur_server_ids:
             self.cur_server_ids[index_id] = random.randint(0, self.num_indexes - 1)
         cur_server_id = self.cur_server_ids[index_id]
@@ -207,10 +263,73 @@ class IndexClient:
     ) -> Tuple[np.ndarray, List]:
         q_size = query.shape[0]
         maximize_metric: bool = self.cfg.metric == "dot"
+        if self.dist_mode:
+            return self._dist_search(query, topk, index_id, maximize_metric,
+                                     return_embeddings)
         results = self.pool.imap(
             lambda idx: idx.search(index_id, query, topk, return_embeddings), self.sub_indexes
         )
         return self._aggregate_results(results, topk, q_size, maximize_metric, return_embeddings)
+
+    def _dist_search(self, query, topk, index_id, maximize_metric,
+                     return_embeddings):
+        """One-rank-per-shard fan-out: local search -> all-gather of
+        (D, I) top-k over RCCL/gloo -> k-way merge (dfann_merge_topk on
+        GPU) -> metadata mapped host-side from (shard, slot). Merge
+        semantics identical to _aggregate_results (incl. quirk-2 dot
+        negation and FLT_MAX pads that can win); metadata rides a
+        collective object gather (the reference ships it in the RPC
+        responses, ref client.py:277-281)."""
+        import torch
+
+        from .dist import (
+            all_gather_object,
+            allgather_shard_topk,
+            merge_gathered_full,
+        )
+
+        scores, ids, meta, embs = self.sub_indexes[0].search_full(
+            index_id, query, topk, return_embeddings)
+        dev = "cuda" if torch.cuda.is_available() else "cpu"
+        D = torch.as_tensor(np.ascontiguousarray(scores, dtype=np.float32),
+                            device=dev)
+        I = torch.as_tensor(np.ascontiguousarray(ids, dtype=np.int64),
+                            device=dev)
+        Dall, Iall = allgather_shard_topk(D, I)
+        Dm, s_idx, j_idx, _local = merge_gathered_full(
+            Dall, Iall, topk, maximize_metric)
+        meta_all = all_gather_object(meta)
+        nq = Dm.shape[0]
+        merged_meta = [
+            [meta_all[s_idx[i, j]][i][j_idx[i, j]] for j in range(topk)]
+            for i in range(nq)
+        ]
+        if return_embeddings:
+            embs_all = all_gather_object(
+                np.asarray(embs, dtype=np.float32) if embs is not None else None)
+            merged_embs = np.stack([
+                np.stack([embs_all[s_idx[i, j]][i][j_idx[i, j]]
+                          for j in range(topk)])
+                for i in range(nq)
+            ])
+            return Dm, merged_meta, merged_embs
+        return Dm, merged_meta
+
+    def search_dev(self, qt, topk: int, index_id: str):
+        """Device-resident serving step — the SAME pipeline search() uses
+        in dist mode, minus the host metadata epilogue: local shard
+        search -> RCCL all-gather -> on-GPU merge. Returns cuda tensors
+        (D (nq,k) f32 with reference sign conventions incl. quirk-2
+        negation, shard_idx (nq,k) i64, local_ids (nq,k) i64). This is
+        the bench's timed region (HIP-graph capturable)."""
+        from .dist import allgather_shard_topk, merge_gathered_full
+
+        maximize_metric: bool = self.cfg.metric == "dot"
+        D, I = self.sub_indexes[0].search_ids_dev(index_id, qt, topk)
+        Dall, Iall = allgather_shard_topk(D, I)
+        Dm, s_idx, _j, local = merge_gathered_full(
+            Dall, Iall, topk, maximize_metric, device_out=True)
+        return Dm, s_idx, local
 
     def search_with_filter(
         self,
@@ -310,23 +429,44 @@ class IndexClient:
     # -- introspection / knobs --------------------------------------------
 
     def get_centroids(self, index_id: str):
+        if self.dist_mode:
+            from .dist import all_gather_object
+
+            return all_gather_object(
+                self.sub_indexes[0].get_centroids(index_id))
         return self.pool.map(lambda idx: idx.get_centroids(index_id), self.sub_indexes)
 
     def set_nprobe(self, index_id: str, nprobe: int):
         return self.pool.map(lambda idx: idx.set_nprobe(index_id, nprobe), self.sub_indexes)
 
     def get_state(self, index_id: str) -> IndexState:
-        states = self.pool.map(lambda idx: idx.get_state(index_id), self.sub_indexes)
+        if self.dist_mode:
+            from .dist import all_gather_object
+
+            states = all_gather_object(self.sub_indexes[0].get_state(index_id))
+        else:
+            states = self.pool.map(lambda idx: idx.get_state(index_id), self.sub_indexes)
         return IndexState.get_aggregated_states(states)
 
     def add_buffer_to_index(self, index_id: str):
         self.pool.map(lambda idx: idx.add_buffer_to_index(index_id), self.sub_indexes)
 
     def get_ntotal(self, index_id: str) -> int:
+        if self.dist_mode:
+            from .dist import all_gather_object
+
+            return sum(all_gather_object(
+                self.sub_indexes[0].get_ntotal(index_id)))
         return sum(self.pool.map(lambda idx: idx.get_ntotal(index_id), self.sub_indexes))
 
     def get_ids(self, index_id: str) -> set:
-        id_set_list = self.pool.map(lambda idx: idx.get_ids(index_id), self.sub_indexes)
+        if self.dist_mode:
+            from .dist import all_gather_object
+
+            id_set_list = all_gather_object(
+                self.sub_indexes[0].get_ids(index_id))
+        else:
+            id_set_list = self.pool.map(lambda idx: idx.get_ids(index_id), self.sub_indexes)
         return set().union(*id_set_list)
 
     def set_omp_num_threads(self, num_threads: int) -> None:
@@ -340,4 +480,4 @@ class IndexClient:
                 close_fn()
 
     def get_num_servers(self):
-        return self.num_indexes
+        return self.num_shards

@@ -54,14 +54,16 @@ def allgather_shard_topk(D_local, I_local):
     return torch.stack(Dg), torch.stack(Ig)
 
 
-def merge_gathered(Dall, Iall, k, maximize, device_out=False):
+def merge_gathered_full(Dall, Iall, k, maximize, device_out=False):
     """Merge (S,nq,k) shard results into (nq,k), reference heap semantics
     (returned distances NEGATED for maximize — quirk 2). Returns
-    (D, shard_idx, local_ids): shard_idx/local_ids map each winner back to
-    its shard and per-shard id, mirroring client.py:290,297-298.
-    device_out=True keeps the three results in HBM (serving step: no
-    per-step D2H sync; graph-capturable) — the reference-API client path
-    always returns host arrays."""
+    (D, shard_idx, slot_idx, local_ids): shard_idx/slot_idx locate each
+    winner in the gathered (S,nq,k) input (slot_idx = j within the
+    shard's k row — what the metadata map needs), local_ids are the
+    winner's shard-local ids, mirroring client.py:290,297-298.
+    device_out=True keeps the results in HBM (serving step: no per-step
+    D2H sync; graph-capturable) — the reference-API client path always
+    returns host arrays."""
     if Dall.is_cuda:
         import torch
 
@@ -72,10 +74,12 @@ def merge_gathered(Dall, Iall, k, maximize, device_out=False):
         # slot = s*nq*k + q*k + j IS the flat index into Iall — decode on
         # the GPU (host numpy here cost ~half the step time at 10k batch)
         s_idx = torch.div(slots, nq * kk, rounding_mode="floor")
+        j_idx = slots % kk
         local = Iall.reshape(-1)[slots]
         if device_out:
-            return Dm, s_idx, local
-        return Dm.cpu().numpy(), s_idx.cpu().numpy(), local.cpu().numpy()
+            return Dm, s_idx, j_idx, local
+        return (Dm.cpu().numpy(), s_idx.cpu().numpy(), j_idx.cpu().numpy(),
+                local.cpu().numpy())
     # CPU (gloo tests): numpy restatement
     Da = Dall.numpy()
     Ia = Iall.numpy()
@@ -85,10 +89,33 @@ def merge_gathered(Dall, Iall, k, maximize, device_out=False):
     slots = np.arange(S * kk)
     Dout = np.empty((nq, k), dtype=np.float32)
     s_out = np.empty((nq, k), dtype=np.int64)
+    j_out = np.empty((nq, k), dtype=np.int64)
     l_out = np.empty((nq, k), dtype=np.int64)
     for i in range(nq):
         order = np.lexsort((slots, flat[i]))[:k]
         Dout[i] = flat[i, order]
         s_out[i] = order // kk
+        j_out[i] = order % kk
         l_out[i] = Ia[order // kk, i, order % kk]
-    return Dout, s_out, l_out
+    return Dout, s_out, j_out, l_out
+
+
+def merge_gathered(Dall, Iall, k, maximize, device_out=False):
+    """(D, shard_idx, local_ids) — see merge_gathered_full."""
+    D, s_idx, _j, local = merge_gathered_full(Dall, Iall, k, maximize,
+                                              device_out=device_out)
+    return D, s_idx, local
+
+
+def all_gather_object(obj):
+    """All-gather an arbitrary picklable object -> list indexed by rank
+    (shard order). Single-process: [obj]. Used for the metadata halves of
+    the dist client search (the reference ships metadata over its RPC
+    responses, client.py:277-281; here it rides torch.distributed)."""
+    import torch.distributed as dist
+
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return [obj]
+    out = [None] * dist.get_world_size()
+    dist.all_gather_object(out, obj)
+    return out

@@ -117,12 +117,35 @@ def _factory_spec(cfg: IndexCfg, total_data_size: int) -> dict:
     return spec
 
 
+# Engine performance knobs passed through from cfg.extra (our extension:
+# the reference has no equivalent — faiss knobs ride the factory string).
+# Documented in INTEGRATION.md; consumed by the HIP engine
+# (csrc/dfann.hip create_from_spec), ignored by the test oracle.
+ENGINE_PERF_KNOBS = (
+    "coarse_bf16",    # assign/coarse GEMMs on bf16 MFMA (approximate path)
+    "max_ppc",        # k-means subsample cap per centroid
+    "ws_mb",          # workspace chunk budget (key matrices)
+    "pq_precomputed", # PQ-L2 term2/term3 tables
+    "pq_lut_global",  # ADC LUTs built to HBM (GLUT scan path)
+    "pq_lut_mb",      # GLUT chunk budget
+    "pq_lut_f16",     # fp16 ADC tables (tolerance path)
+    "scan_fan",       # list-segment fan (experiment knob)
+)
+
+
+def _apply_perf_knobs(cfg: IndexCfg, spec: dict) -> dict:
+    for k in ENGINE_PERF_KNOBS:
+        if k in cfg.extra:
+            spec[k] = cfg.extra[k]
+    return spec
+
+
 def resolve_engine_spec(cfg: IndexCfg, total_data_size: int) -> dict:
     """Mirror of reference Index._init_faiss_index (index.py:380-401)."""
     if cfg.index_builder_type:
-        return _builder_spec(cfg)
+        return _apply_perf_knobs(cfg, _builder_spec(cfg))
     if cfg.faiss_factory:
-        return _factory_spec(cfg, total_data_size)
+        return _apply_perf_knobs(cfg, _factory_spec(cfg, total_data_size))
     raise RuntimeError(
         "Either faiss_factory or valid index_builder_type should be specified "
         "to initialize index"

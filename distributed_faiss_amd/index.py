@@ -169,6 +169,17 @@ class Index:
     def search(
         self, query_batch: np.ndarray, top_k: int = 100, return_embeddings: bool = False
     ) -> Tuple[np.ndarray, List[List[object]], Optional[np.ndarray]]:
+        scores, _indexes, results_meta, embs = self.search_full(
+            query_batch, top_k, return_embeddings)
+        return scores, results_meta, embs
+
+    def search_full(
+        self, query_batch: np.ndarray, top_k: int = 100, return_embeddings: bool = False
+    ) -> Tuple[np.ndarray, np.ndarray, List[List[object]], Optional[np.ndarray]]:
+        """search() plus the shard-local ids — the distributed client's
+        fan-out needs ids alongside metadata so the RCCL-gathered merge
+        can map winners back (ref client.py:290,297-298 does this with
+        synthetic ids; here the (shard, slot) pair does it)."""
         query_batch = np.ascontiguousarray(query_batch, dtype=np.float32)
         with self.index_lock:
             if self.state != IndexState.TRAINED:
@@ -179,16 +190,29 @@ class Index:
                 scores, indexes = self.engine.search(query_batch, top_k)
                 embs = None
 
+        return scores, indexes, self.metadata_for(indexes), embs
+
+    def metadata_for(self, indexes: np.ndarray) -> List[List[object]]:
+        """Map an (nq, k) id matrix to metadata (-1 -> None), the gather
+        the reference does inline at index.py:182-191."""
         nq, n = indexes.shape
         with self.buffer_lock:
-            results_meta = [
+            return [
                 [
                     self.id_to_metadata[indexes[i, j]] if indexes[i, j] != -1 else None
                     for j in range(n)
                 ]
                 for i in range(nq)
             ]
-        return scores, results_meta, embs
+
+    def search_ids_dev(self, qt, top_k: int):
+        """Device-resident search: (D, I) stay in HBM (torch cuda
+        tensors). Serving-step hot path for the distributed client — no
+        metadata epilogue, no D2H. HIP engine only."""
+        with self.index_lock:
+            if self.state != IndexState.TRAINED:
+                raise RuntimeError(f"Server index is not trained. state: {self.state}")
+            return self.engine.search_dev(qt, top_k)
 
     # -- persistence -------------------------------------------------------
 

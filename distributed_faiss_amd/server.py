@@ -147,6 +147,27 @@ class IndexServer:
         index = self._get_index(index_id)
         return index.search(query_batch, top_k=top_k, return_embeddings=return_embeddings)
 
+    def search_full(
+        self, index_id: str, query_batch: np.ndarray, top_k: int,
+        return_embeddings: bool = False
+    ) -> Tuple:
+        """search() plus shard-local ids (distributed-client fan-out)."""
+        index = self._get_index(index_id)
+        return index.search_full(query_batch, top_k=top_k,
+                                 return_embeddings=return_embeddings)
+
+    def search_ids_dev(self, index_id: str, qt, top_k: int):
+        """Device-resident (D, I) search — the distributed client's timed
+        serving step (no metadata epilogue, results stay in HBM)."""
+        return self._get_index(index_id).search_ids_dev(qt, top_k)
+
+    def adopt_index(self, index_id: str, index) -> None:
+        """Install an externally built Index under an id (harness
+        plumbing: bench.py builds its shard engine from device-resident
+        synthetic data, then serves it through the reference surface)."""
+        with self.indexes_lock:
+            self.indexes[index_id] = index
+
     def sync_train(self, index_id: str):
         self._get_index(index_id).train()
 
