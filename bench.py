@@ -179,12 +179,17 @@ def main():
 
     import torch
 
+    from distributed_faiss_amd.client import IndexClient
     from distributed_faiss_amd.dist import (
         allgather_shard_topk,
         init_from_env,
         merge_gathered,
     )
-    from distributed_faiss_amd.hip_engine import HipEngine
+    from distributed_faiss_amd.hip_engine import HipEngine, HipProvider
+    from distributed_faiss_amd.index import Index
+    from distributed_faiss_amd.index_cfg import IndexCfg
+    from distributed_faiss_amd.index_state import IndexState
+    from distributed_faiss_amd.server import IndexServer
 
     rank, world = init_from_env()
     assert world == args.gpus or world == 1, (world, args.gpus)
@@ -229,6 +234,33 @@ def main():
     torch.cuda.synchronize()
     log(f"added {eng.ntotal} in {time.time()-t0:.1f}s")
 
+    # ---- serve through the reference surface (IndexClient) ----
+    # The timed step routes through IndexClient.search_dev — the SAME
+    # pipeline IndexClient.search uses in distributed mode (local shard
+    # search -> RCCL all-gather -> on-GPU k-way merge), minus the host
+    # metadata epilogue. The shard engine was built above from
+    # device-resident synthetic data (the Index.add_batch path would
+    # bounce the 38 GB shard through host numpy), then adopted into the
+    # server, exactly what Index.from_storage_dir does on load.
+    class _NoMeta:  # bench shards carry no metadata; every id maps to None
+        def __getitem__(self, i):
+            return None
+
+        def __len__(self):
+            return 1 << 62
+
+    icfg = IndexCfg(faiss_factory="Flat", dim=cfg["d"],
+                    metric=("l2" if metric == 1 else "dot"), nprobe=1)
+    shard_index = Index(icfg, provider=HipProvider())
+    shard_index.engine = eng
+    shard_index.state = IndexState.TRAINED
+    shard_index.id_to_metadata = _NoMeta()
+    server = IndexServer(rank, os.path.join("/tmp", "dfann_bench"))
+    server.adopt_index("bench", shard_index)
+    client = IndexClient(servers=[server])
+    client.cfg = icfg
+    assert client.dist_mode == (world > 1)
+
     # ---- ground truth + nprobe operating point ----
     t0 = time.time()
     Dgt, Igt = exact_ground_truth(xb, q, metric, k, device)
@@ -238,12 +270,11 @@ def main():
     log(f"ground truth in {time.time()-t0:.1f}s")
 
     def run_search(nprobe, qt):
-        # serving step: results stay in HBM (no per-step D2H sync; the
-        # reference-API client path still returns host arrays)
+        # serving step THROUGH the client surface: results stay in HBM
+        # (no per-step D2H sync; IndexClient.search wraps this same
+        # pipeline and adds the host metadata epilogue)
         eng.nprobe = nprobe
-        D, I = eng.search_dev(qt, k)
-        Da, Ia = allgather_shard_topk(D, I)
-        Dm, s_i, loc = merge_gathered(Da, Ia, k, maximize, device_out=True)
+        Dm, s_i, loc = client.search_dev(qt, k, "bench")
         return Dm, s_i * cfg["n"] + loc
 
     def recall_at(nprobe, nq_eval=2048):
@@ -269,6 +300,16 @@ def main():
             nprobe, recall = cand, r
     eng.nprobe = nprobe
     log(f"operating point: nprobe={nprobe} recall={recall:.4f}")
+
+    # one untimed pass through the FULL reference API (host arrays +
+    # metadata epilogue) to prove the served path and the timed path
+    # agree — client.search wraps the same pipeline as search_dev
+    q4 = q[:4].cpu().numpy()
+    Dh, _meta_h = (client.search(q4, k, "bench")[:2])
+    Dd, _gid = run_search(nprobe, q[:4].contiguous())
+    # both carry the reference merge conventions (incl. quirk-2 negation)
+    assert np.allclose(np.asarray(Dh), Dd.cpu().numpy(), rtol=1e-6, atol=1e-6), \
+        "client.search diverged from the device serving step"
 
     # ---- timed region ----
     import torch.distributed as tdist
@@ -465,6 +506,7 @@ def main():
                                      else args.pq_lut_f16)
                            else "f32"),
                 "step_graph": bool(use_graph),
+                "serving_path": "IndexClient.search_dev",
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
