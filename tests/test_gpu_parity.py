@@ -585,3 +585,46 @@ def test_client_server_sharded_equals_single_gpu(tmp_path):
     Ds, Ms = sc.search(q, 5, "x")
     assert (Da == Ds).all()
     assert Ma == Ms
+
+
+def test_client_search_dev_matches_search(tmp_path):
+    # the bench's timed step (IndexClient.search_dev: device-resident
+    # local search -> all-gather -> dfann_merge_topk) must agree with the
+    # full reference API (IndexClient.search: host arrays + metadata),
+    # which wraps the same pipeline — incl. the quirk-2 dot negation
+    from distributed_faiss_amd import IndexCfg, IndexClient, IndexServer
+
+    for metric_name in ("dot", "l2"):
+        prov = HipProvider()
+        srv = IndexServer(0, str(tmp_path / metric_name), provider=prov)
+        cli = IndexClient(servers=[srv])
+        cfg = (IndexCfg(index_builder_type="flat", dim=32, metric="dot")
+               if metric_name == "dot"
+               else IndexCfg(faiss_factory="Flat", dim=32, metric="l2"))
+        cli.create_index("x", cfg)
+        cli.cfg = cfg
+        rng = np.random.default_rng(31)
+        emb = rng.random((800, 32), dtype=np.float32)
+        cli.add_index_data("x", emb, [("m", i) for i in range(800)],
+                           train_async_if_triggered=False)
+        cli.sync_train("x")
+        import time
+
+        from distributed_faiss_amd import IndexState
+
+        for _ in range(200):
+            if (cli.get_state("x") == IndexState.TRAINED
+                    and cli.get_ntotal("x") == 800):
+                break
+            time.sleep(0.05)
+        q = rng.random((6, 32), dtype=np.float32)
+        Dh, Mh = cli.search(q, 5, "x")
+        qt = torch.as_tensor(q).cuda()
+        Dd, s_idx, local = cli.search_dev(qt, 5, "x")
+        np.testing.assert_allclose(np.asarray(Dh), Dd.cpu().numpy(),
+                                   rtol=1e-6, atol=1e-6)
+        # winners map to the same metadata: shard 0, id -> ("m", id)
+        loc = local.cpu().numpy()
+        for i in range(6):
+            for j in range(5):
+                assert Mh[i][j] == ("m", int(loc[i, j]))
