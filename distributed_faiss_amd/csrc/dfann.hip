@@ -229,11 +229,11 @@ struct dfann_index {
 
   // timing
   bool timing = false;
-  std::vector<TimingEv> ev_scan, ev_gemm, ev_merge;
+  std::vector<TimingEv> ev_scan, ev_gemm, ev_merge, ev_lut;
   int64_t scan_rows = 0, scan_bytes = 0, gemm_flops = 0;
 
   ~dfann_index() {
-    for (auto &v : {ev_scan, ev_gemm, ev_merge})
+    for (auto &v : {ev_scan, ev_gemm, ev_merge, ev_lut})
       for (auto &e : v) {
         (void)hipEventDestroy(e.a);
         (void)hipEventDestroy(e.b);
@@ -849,15 +849,18 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   cand_d = h->ws3.as<float>();
   cand_p = h->ws4.as<unsigned>();
   TimingEv e;
-  if (h->timing) e = h->ev_begin(stream);
   if (use_pre) {
-    // per-batch tables: q norms + term3
+    // per-batch tables: q norms + term3 (counted as LUT-build time)
+    TimingEv el;
+    if (h->timing) el = h->ev_begin(stream);
     h->qn_ws.ensure((size_t)nq * 4);
     h->term3_ws.ensure((size_t)nq * h->m * 256 * 4);
     rownorms(q, nq, h->d, h->qn_ws.as<float>(), stream);
     hipLaunchKernelGGL(k_pq_term3, grid1d(nq * (int64_t)h->m * 256), dim3(256),
                        0, stream, q, h->codebooks.as<float>(), nq, h->m,
                        h->dsub, h->term3_ws.as<float>());
+    if (h->timing) h->ev_end(el, stream, h->ev_lut);
+    if (h->timing) e = h->ev_begin(stream);
     auto pk = rk ? k_scan_pq_l2_pre_rk : k_scan_pq_l2_pre;
     hipLaunchKernelGGL(pk, dim3((unsigned)(nq * nprobe)), dim3(scan_bs), lds,
                        stream, q, h->centroids.as<float>(),
@@ -867,6 +870,7 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
                        cand_d, cand_p, fam_floats, h->term2.as<float>(),
                        h->term3_ws.as<float>(), h->qn_ws.as<float>());
+    if (h->timing) h->ev_end(e, stream, h->ev_scan);
   } else if (use_glut) {
     auto gk = lut_f16 ? (rk ? (ip ? k_scan_pq_ip_gh_rk : k_scan_pq_l2_gh_rk)
                             : (ip ? k_scan_pq_ip_gh : k_scan_pq_l2_gh))
@@ -892,6 +896,8 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
       long long qpn = (long long)nqc * nprobe;
       dim3 lg((unsigned)((qpn + PQ_LUT_QPT - 1) / PQ_LUT_QPT),
               (unsigned)h->m);
+      TimingEv el;
+      if (h->timing) el = h->ev_begin(stream);
       if (lut_f16) {
         hipLaunchKernelGGL(k_pq_lut_f16, lg, dim3(256), lut_lds, stream,
                            q + q0 * h->d, h->centroids.as<float>(),
@@ -905,6 +911,8 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                            (int)nqc, nprobe, h->d, h->m, h->dsub, ip ? 1 : 0,
                            lutg);
       }
+      if (h->timing) h->ev_end(el, stream, h->ev_lut);
+      if (h->timing) e = h->ev_begin(stream);
       hipLaunchKernelGGL(gk, dim3((unsigned)(nqc * nprobe)), dim3(scan_bs),
                          lds, stream, q + q0 * h->d, h->centroids.as<float>(),
                          h->codebooks.as<float>(), h->sq_vmin.as<float>(),
@@ -914,8 +922,10 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                          h->m, h->dsub, k, h->stride,
                          cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
                          fam_floats, lutg);
+      if (h->timing) h->ev_end(e, stream, h->ev_scan);
     }
   } else {
+    if (h->timing) e = h->ev_begin(stream);
     hipLaunchKernelGGL(kern, dim3((unsigned)(nq * nprobe * fan)),
                        dim3(scan_bs), lds, stream, q, h->centroids.as<float>(),
                        h->codebooks.as<float>(), h->sq_vmin.as<float>(),
@@ -923,9 +933,9 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
                        (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
                        cand_d, cand_p, fam_floats, fan);
+    if (h->timing) h->ev_end(e, stream, h->ev_scan);
   }
   if (h->timing) {
-    h->ev_end(e, stream, h->ev_scan);
     // algorithmic units: sum of probed list lengths
     std::vector<int32_t> hp((size_t)nq * nprobe);
     HIP_CHECK(hipMemcpy(hp.data(), probes, hp.size() * 4, hipMemcpyDeviceToHost));
@@ -1437,9 +1447,11 @@ extern "C" int dfann_get_timing(dfann_index *h, dfann_timing *out) {
   API_BEGIN
   out->scan_launches = (int64_t)h->ev_scan.size();
   out->merge_launches = (int64_t)h->ev_merge.size();
+  out->lut_launches = (int64_t)h->ev_lut.size();
   out->scan_ms = sum_events(h->ev_scan);
   out->gemm_ms = sum_events(h->ev_gemm);
   out->merge_ms = sum_events(h->ev_merge);
+  out->lut_ms = sum_events(h->ev_lut);
   out->scan_rows = h->scan_rows;
   out->scan_bytes = h->scan_bytes;
   out->gemm_flops = h->gemm_flops;

@@ -29,6 +29,8 @@ class DfannTiming(ctypes.Structure):
         ("gemm_flops", ctypes.c_int64),
         ("merge_ms", ctypes.c_double),
         ("merge_launches", ctypes.c_int64),
+        ("lut_ms", ctypes.c_double),
+        ("lut_launches", ctypes.c_int64),
     ]
 
 

@@ -147,7 +147,7 @@ int dfann_merge_topk(int64_t nq, int S, int k, const float *D_dev,
 /* --- kernel timing for the roofline harness (bench.py) ---------------- */
 
 typedef struct dfann_timing {
-  double scan_ms;      /* ivf list-scan kernel, summed over launches */
+  double scan_ms;      /* ivf list-scan kernel ALONE, summed over launches */
   int64_t scan_launches;
   int64_t scan_rows;   /* codes scanned (algorithmic units) */
   int64_t scan_bytes;  /* scan_rows * packed code stride */
@@ -155,6 +155,8 @@ typedef struct dfann_timing {
   int64_t gemm_flops;  /* 2*M*N*K summed */
   double merge_ms;
   int64_t merge_launches;
+  double lut_ms;       /* ADC-table build (k_pq_lut / term3), GLUT/PRE paths */
+  int64_t lut_launches;
 } dfann_timing;
 
 int dfann_set_timing(dfann_index *h, int enabled);
