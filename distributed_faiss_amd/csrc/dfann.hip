@@ -913,15 +913,53 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
       }
       if (h->timing) h->ev_end(el, stream, h->ev_lut);
       if (h->timing) e = h->ev_begin(stream);
-      hipLaunchKernelGGL(gk, dim3((unsigned)(nqc * nprobe)), dim3(scan_bs),
-                         lds, stream, q + q0 * h->d, h->centroids.as<float>(),
-                         h->codebooks.as<float>(), h->sq_vmin.as<float>(),
-                         h->sq_scale.as<float>(), probes + q0 * nprobe,
-                         keys + q0 * nprobe, h->cr_codes.as<uint8_t>(),
-                         h->cr_off.as<int64_t>(), (int)nqc, nprobe, h->d,
-                         h->m, h->dsub, k, h->stride,
-                         cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
-                         fam_floats, lutg);
+      // persistent multi-pair variant (fp16 LUT + register top-k): a
+      // fixed grid strides over the (query, probe) pairs and prefetches
+      // the next pair's LUT during the current scan — the one-pair
+      // kernel is staging-dominated at short lists (see kernels.hip).
+      // Bit-identical results; DFANN_SCAN_PERS=0 forces the old path.
+      int pfn = 0;
+      long long qpn_c = (long long)nqc * nprobe;
+      if (lut_f16 && rk) {
+        int lut_u4 = h->m * 32;  // m*256 halves / 8 per uint4
+        if (lut_u4 % (int)scan_bs == 0) {
+          int cand = lut_u4 / (int)scan_bs;
+          if (cand == 2 || cand == 4 || cand == 8 || cand == 16) pfn = cand;
+        }
+        if (const char *e2 = getenv("DFANN_SCAN_PERS"))
+          if (atoi(e2) == 0) pfn = 0;
+      }
+      if (pfn) {
+        auto pk = ip ? (pfn == 2 ? k_scan_pq_ip_ghp2
+                        : pfn == 4 ? k_scan_pq_ip_ghp4
+                        : pfn == 8 ? k_scan_pq_ip_ghp8 : k_scan_pq_ip_ghp16)
+                     : (pfn == 2 ? k_scan_pq_l2_ghp2
+                        : pfn == 4 ? k_scan_pq_l2_ghp4
+                        : pfn == 8 ? k_scan_pq_l2_ghp8 : k_scan_pq_l2_ghp16);
+        long long pb = 2048;
+        if (const char *e2 = getenv("DFANN_SCAN_PERS_BLOCKS"))
+          if (atoll(e2) > 0) pb = atoll(e2);
+        if (pb > qpn_c) pb = qpn_c;
+        hipLaunchKernelGGL(pk, dim3((unsigned)pb), dim3(scan_bs), lds, stream,
+                           q + q0 * h->d, h->centroids.as<float>(),
+                           h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                           h->sq_scale.as<float>(), probes + q0 * nprobe,
+                           keys + q0 * nprobe, h->cr_codes.as<uint8_t>(),
+                           h->cr_off.as<int64_t>(), (int)nqc, nprobe, h->d,
+                           h->m, h->dsub, k, h->stride,
+                           cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
+                           fam_floats, lutg);
+      } else {
+        hipLaunchKernelGGL(gk, dim3((unsigned)(nqc * nprobe)), dim3(scan_bs),
+                           lds, stream, q + q0 * h->d, h->centroids.as<float>(),
+                           h->codebooks.as<float>(), h->sq_vmin.as<float>(),
+                           h->sq_scale.as<float>(), probes + q0 * nprobe,
+                           keys + q0 * nprobe, h->cr_codes.as<uint8_t>(),
+                           h->cr_off.as<int64_t>(), (int)nqc, nprobe, h->d,
+                           h->m, h->dsub, k, h->stride,
+                           cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
+                           fam_floats, lutg);
+      }
       if (h->timing) h->ev_end(e, stream, h->ev_scan);
     }
   } else {

@@ -1430,6 +1430,127 @@ INSTANTIATE_SCAN_GLUT_H(k_scan_pq_l2_gh_rk, false, true)
 INSTANTIATE_SCAN_GLUT_H(k_scan_pq_ip_gh_rk, true, true)
 
 // ---------------------------------------------------------------------------
+// PERSISTENT fp16-GLUT scan (k <= 16 register selection): a fixed grid of
+// blocks loops over (query, probe) pairs with stride gridDim.x, and each
+// block PREFETCHES the NEXT pair's ADC table into registers while it
+// scans the current pair. Motivation (profiles/r01/pmc_sq_waits_1m.csv +
+// r1 ladder): at the headline shape a pair's list is ~190 rows — the
+// one-pair-per-block kernel is STAGING-dominated (stage 32 KB LUT, scan
+// ~12 KB of codes, die) and phase-converged (53% of wave cycles parked).
+// Here the LUT round-trip latency hides under the previous pair's scan +
+// top-k extraction, and block phases de-correlate naturally.
+//   * values are BIT-IDENTICAL to the one-pair kernel: same LUT bytes,
+//     same per-row accumulation order, same (dist, pos) selection.
+//   * PFN = LUT uint4 groups per thread (lut_bytes / blockDim / 16);
+//     host dispatches among the instantiations below and falls back to
+//     the one-pair kernel when the shape doesn't divide.
+//   * the RegTopK extraction scratch aliases the LDS LUT (as in
+//     ivf_scan_body); the prefetched registers rewrite the full LUT for
+//     the next pair after extraction's trailing barrier.
+// ---------------------------------------------------------------------------
+
+template <bool IS_IP, int PFN>
+__device__ void scan_pq_ghp_body(
+    const float *__restrict__ q, const float *__restrict__ cent,
+    const float *__restrict__ cb, const float *__restrict__ sq_vmin,
+    const float *__restrict__ sq_scale, const int *__restrict__ probes,
+    const float *__restrict__ keys, const uint8_t *__restrict__ codes,
+    const int64_t *__restrict__ off, int nq, int nprobe, int d, int m,
+    int dsub, int k, int stride, float *__restrict__ cand_d,
+    unsigned *__restrict__ cand_p, int fam_floats,
+    const float *__restrict__ glut) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float *fam = reinterpret_cast<float *>(smem);
+  uint4 *lds4 = reinterpret_cast<uint4 *>(smem);
+  const int BS = blockDim.x;
+  long long qpn = (long long)nq * nprobe;
+  const int lut_u4 = PFN * BS;  // uint4 groups in one pair's fp16 LUT
+  const uint4 *lutg = reinterpret_cast<const uint4 *>(glut);
+
+  long long pair = blockIdx.x;
+  if (pair >= qpn) return;
+  // stage the first pair's LUT via global->LDS DMA
+  {
+    const uint4 *src = lutg + pair * (size_t)lut_u4;
+    for (int e = threadIdx.x; e < lut_u4; e += BS)
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(src + e),
+          (__attribute__((address_space(3))) unsigned int *)(lds4 + e), 16, 0,
+          0);
+  }
+  uint4 pf[PFN];
+  for (; pair < qpn; pair += gridDim.x) {
+    long long nxt = pair + gridDim.x;
+    __syncthreads();  // staged/rewritten LUT visible (drains the glds)
+    // prefetch the next pair's LUT into registers: these loads fly
+    // together with the scan's row loads below
+    if (nxt < qpn) {
+      const uint4 *src = lutg + nxt * (size_t)lut_u4;
+#pragma unroll
+      for (int i = 0; i < PFN; ++i) pf[i] = src[i * BS + threadIdx.x];
+    }
+    // ---- scan this pair (same order/ops as ivf_scan_body REGSEL) ----
+    int L = probes[pair];
+    long long out_base = pair * k;
+    long long s0 = off[L], s1 = off[L + 1];
+    float bias = 0.f;
+    if (IS_IP) bias = -keys[pair];
+    if (s0 == s1) {
+      for (int j = threadIdx.x; j < k; j += BS) {
+        cand_d[out_base + j] = DFANN_FLT_MAX;
+        cand_p[out_base + j] = PAD_POS;
+      }
+    } else {
+      RegTopK<16> loc;
+      loc.init();
+      for (long long base = s0; base < s1; base += (long long)4 * BS) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          long long pos = base + (long long)u * BS + threadIdx.x;
+          if (pos < s1) {
+            const uint8_t *cp = codes + pos * (size_t)stride;
+            float acc = scan_row_dist<0, IS_IP, true>(cp, fam, d, m);
+            float dist = IS_IP ? -(bias + acc) : acc;
+            loc.push(dist, (unsigned)pos);
+          }
+        }
+      }
+      __syncthreads();  // all lanes done reading the LDS LUT
+      regtopk_block_extract<16>(loc, k, smem, cand_d + out_base,
+                                cand_p + out_base);
+    }
+    // install the prefetched LUT (extraction's trailing barrier ensures
+    // no lane still reads the old one; empty pairs never read it)
+    if (nxt < qpn) {
+      __syncthreads();  // pad-write case has no extraction barrier
+#pragma unroll
+      for (int i = 0; i < PFN; ++i) lds4[i * BS + threadIdx.x] = pf[i];
+    }
+  }
+}
+
+#define INSTANTIATE_SCAN_GHP(NAME, IS_IP, PFN)                                 \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
+      unsigned *cand_p, int fam_floats, const float *glut) {                   \
+    scan_pq_ghp_body<IS_IP, PFN>(q, cent, cb, sq_vmin, sq_scale, probes,       \
+                                 keys, codes, off, nq, nprobe, d, m, dsub, k,  \
+                                 stride, cand_d, cand_p, fam_floats, glut);    \
+  }
+
+INSTANTIATE_SCAN_GHP(k_scan_pq_l2_ghp2, false, 2)
+INSTANTIATE_SCAN_GHP(k_scan_pq_ip_ghp2, true, 2)
+INSTANTIATE_SCAN_GHP(k_scan_pq_l2_ghp4, false, 4)
+INSTANTIATE_SCAN_GHP(k_scan_pq_ip_ghp4, true, 4)
+INSTANTIATE_SCAN_GHP(k_scan_pq_l2_ghp8, false, 8)
+INSTANTIATE_SCAN_GHP(k_scan_pq_ip_ghp8, true, 8)
+INSTANTIATE_SCAN_GHP(k_scan_pq_l2_ghp16, false, 16)
+INSTANTIATE_SCAN_GHP(k_scan_pq_ip_ghp16, true, 16)
+
+// ---------------------------------------------------------------------------
 // k_pq_lut: ADC lookup tables to HBM, one 256-entry row per (query,
 // probe, subspace) at out[(qp)*m*256 + j*256 + c] — the scan's GLUT
 // staging then reads its (query, probe) block as one contiguous,
