@@ -81,11 +81,14 @@ WORKLOADS = {
         coarse_bf16=1, max_ppc=64,
     ),
     # THE HEADLINE: BASELINE.json configs[3] per-shard slice — 12.5M x 768
-    # IVFPQ m=64, nlist=65536/shard; at --gpus 8 the sharded DB is exactly
-    # the 100M x 768 configuration. bf16-MFMA assign/coarse for the build;
-    # k-means capped at 64 pts/centroid (~4.2M training rows/shard).
+    # IVFPQ m=64, at --gpus 8 the sharded DB is exactly the 100M x 768
+    # configuration. bf16-MFMA assign/coarse for the build; k-means capped
+    # at 64 pts/centroid. nlist=16384/shard (engine tuning, round-2 sweep
+    # gpurun_out/r2s_nl*.json: same scanned rows/query as 65536@nprobe=8
+    # but 4x less coarse GEMM / top-k / ADC-table overhead — 1.34M ->
+    # 3.94M QPS at recall 0.978; override with --nlist).
     "ivfpq_100m8_d768_m64": dict(
-        type="ivfpq", d=768, n=12_500_000, nlist=65536, m=64, nbits=8,
+        type="ivfpq", d=768, n=12_500_000, nlist=16384, m=64, nbits=8,
         metric=1, nq=10_000, k=10, centers=12_500, sigma=0.5, latent=32,
         coarse_bf16=1, max_ppc=64, ws_mb=2048, pq_lut_f16=1,
     ),

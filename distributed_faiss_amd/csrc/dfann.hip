@@ -137,6 +137,101 @@ struct DevBuf {
 };
 
 // ---------------------------------------------------------------------------
+// slab arena: a growable row store made of fixed-size slabs (1<<rlog rows
+// each). The CSR rebuild recycles old slabs behind its write window
+// (rebuild_csr), so code images never need a 2x contiguous transient —
+// the round-1 staging-arena + CSR duplication is gone.
+// Kernels address rows through a device slab-pointer table (kernels.hip
+// slab_row).
+// ---------------------------------------------------------------------------
+
+struct SlabArena {
+  int stride = 0;
+  int rlog = 0;                 // rows per slab = 1 << rlog
+  std::vector<void *> slabs;    // nullptr = freed
+  int64_t rows = 0;             // rows in use (bump)
+  DevBuf table;                 // device copy of `slabs`
+  bool table_dirty = true;
+
+  void init(int stride_, size_t target_slab_bytes = (size_t)512 << 20) {
+    reset();
+    stride = stride_;
+    rlog = 0;
+    while (((size_t)2 << rlog) * stride <= target_slab_bytes) ++rlog;
+  }
+  int64_t rps() const { return (int64_t)1 << rlog; }
+  size_t slab_bytes() const { return (size_t)rps() * stride; }
+  void ensure_rows(int64_t n) {
+    size_t need = (size_t)((n + rps() - 1) >> rlog);
+    while (slabs.size() < need) {
+      void *p = nullptr;
+      HIP_CHECK(hipMalloc(&p, slab_bytes()));
+      slabs.push_back(p);
+      table_dirty = true;
+    }
+  }
+  void free_slab(size_t i) {
+    if (i < slabs.size() && slabs[i]) {
+      (void)hipFree(slabs[i]);
+      slabs[i] = nullptr;  // kernels past the window never touch it
+    }
+  }
+  const uint8_t *const *dev_table(hipStream_t s) {
+    if (table_dirty) {
+      table.ensure(std::max(slabs.size() * 8, (size_t)8));
+      if (!slabs.empty())
+        HIP_CHECK(hipMemcpyAsync(table.p, slabs.data(), slabs.size() * 8,
+                                 hipMemcpyHostToDevice, s));
+      HIP_CHECK(hipStreamSynchronize(s));
+      table_dirty = false;
+    }
+    return table.as<const uint8_t *const>();
+  }
+  uint8_t *const *dev_table_mut(hipStream_t s) {
+    return const_cast<uint8_t *const *>(
+        reinterpret_cast<const uint8_t *const *>(dev_table(s)));
+  }
+  // contiguous-host <-> arena row-range copies (persistence, get_lists)
+  void copy_rows_to_host(void *dst, int64_t row0, int64_t n) const {
+    char *d = (char *)dst;
+    int64_t r = row0;
+    while (n > 0) {
+      int64_t in_slab = std::min<int64_t>(n, rps() - (r & (rps() - 1)));
+      const char *src = (const char *)slabs[r >> rlog] +
+                        (size_t)(r & (rps() - 1)) * stride;
+      HIP_CHECK(hipMemcpy(d, src, (size_t)in_slab * stride,
+                          hipMemcpyDeviceToHost));
+      d += (size_t)in_slab * stride;
+      r += in_slab;
+      n -= in_slab;
+    }
+  }
+  void copy_rows_from_host(const void *src, int64_t row0, int64_t n) {
+    ensure_rows(row0 + n);
+    const char *s = (const char *)src;
+    int64_t r = row0;
+    while (n > 0) {
+      int64_t in_slab = std::min<int64_t>(n, rps() - (r & (rps() - 1)));
+      char *dst = (char *)slabs[r >> rlog] + (size_t)(r & (rps() - 1)) * stride;
+      HIP_CHECK(hipMemcpy(dst, s, (size_t)in_slab * stride,
+                          hipMemcpyHostToDevice));
+      s += (size_t)in_slab * stride;
+      r += in_slab;
+      n -= in_slab;
+    }
+  }
+  void reset() {
+    for (auto *p : slabs)
+      if (p) (void)hipFree(p);
+    slabs.clear();
+    rows = 0;
+    table.free();
+    table_dirty = true;
+  }
+  ~SlabArena() { reset(); }
+};
+
+// ---------------------------------------------------------------------------
 // splitmix64 (identical to oracle/core.py)
 // ---------------------------------------------------------------------------
 
@@ -212,11 +307,15 @@ struct dfann_index {
                              // ranking path for huge nlist
 
   DevBuf centroids, cnorm, codebooks, sq_vmin, sq_vdiff, sq_scale;
-  // staging (arrival order)
-  DevBuf st_codes;
-  std::vector<int32_t> h_assign;
-  // CSR
-  DevBuf cr_codes, cr_ids, cr_off, id2pos;
+  // CSR code image (slab arena; csr_rows rows grouped by list) + pending
+  // appends (slab arena in arrival order, merged in by rebuild_csr)
+  SlabArena csr_arena, pend_arena;
+  std::vector<int32_t> h_pend_assign;  // pending rows' list assignment
+  int64_t csr_rows = 0;                // rows in the CSR image
+  int merge_mb = 2048;  // spec "merge_mb": pending bytes that trigger an
+                        // incremental merge during add (bounds the
+                        // rebuild transient to ~2x this)
+  DevBuf cr_ids, cr_off, id2pos, cr_ids_new;
   std::vector<int64_t> h_off;
   bool dirty = false;
   // flat arena
@@ -533,6 +632,13 @@ static dfann_index *create_from_spec(const std::string &js) {
     h->trained = true;  // flat needs no training
   }
   h->stride = round16(h->code_bytes);
+  h->merge_mb = (int)json_int(js, "merge_mb", 2048);
+  if (h->merge_mb < 1) h->merge_mb = 1;
+  // 64 MB slabs: small enough that an empty index costs little, large
+  // enough that the 1B-row table stays a few thousand L1-hot entries
+  h->csr_arena.init(h->stride, (size_t)64 << 20);
+  h->pend_arena.init(h->stride, (size_t)64 << 20);
+  if (h->nlist > 0) h->h_off.assign(h->nlist + 1, 0);
   return h;
 }
 
@@ -596,6 +702,170 @@ static void train_impl(dfann_index *h, int64_t n, const float *x,
   h->trained = true;
 }
 
+// Merge [old CSR + pending appends] into a fresh CSR slab set.
+// Order-preserving two-source merge: new list l = old rows of l (already
+// in ascending-arrival order) followed by pending rows of l (arrival
+// order via the counting-sort permutation), so CSR position order within
+// a list stays == ascending arrival id (the scan tie-break contract).
+// Old slabs are freed behind the write window: because new pos >= old
+// pos for every old row, old slab s is fully consumed once the write
+// cursor passes (s+1)*rps + pend_rows — peak memory stays
+// ~codes + pending + one slab instead of the round-1 2x duplication.
+static void rebuild_csr(dfann_index *h, hipStream_t stream) {
+  if (!h->dirty) return;
+  int64_t n_pend = (int64_t)h->h_pend_assign.size();
+  int64_t n_old = h->csr_rows;
+  int64_t n_new = n_old + n_pend;
+  // pending counting sort (stable -> arrival order within each list)
+  std::vector<int64_t> pend_off(h->nlist + 1, 0);
+  for (int32_t a : h->h_pend_assign) pend_off[a + 1]++;
+  for (int l = 0; l < h->nlist; ++l) pend_off[l + 1] += pend_off[l];
+  std::vector<unsigned> psrc(std::max<int64_t>(n_pend, 1));
+  {
+    std::vector<int64_t> fill(pend_off.begin(), pend_off.end() - 1);
+    for (int64_t i = 0; i < n_pend; ++i)
+      psrc[fill[h->h_pend_assign[i]]++] = (unsigned)i;
+  }
+  std::vector<int64_t> old_off = h->h_off;
+  if ((int64_t)old_off.size() != h->nlist + 1)
+    old_off.assign(h->nlist + 1, 0);
+  std::vector<int64_t> new_off(h->nlist + 1, 0);
+  for (int l = 0; l < h->nlist; ++l)
+    new_off[l + 1] = new_off[l] + (old_off[l + 1] - old_off[l]) +
+                     (pend_off[l + 1] - pend_off[l]);
+  // device tables
+  DevBuf d_new_off, d_old_off, d_pend_off, d_psrc;
+  d_new_off.ensure((size_t)(h->nlist + 1) * 8);
+  d_old_off.ensure((size_t)(h->nlist + 1) * 8);
+  d_pend_off.ensure((size_t)(h->nlist + 1) * 8);
+  d_psrc.ensure((size_t)std::max<int64_t>(n_pend, 1) * 4);
+  HIP_CHECK(hipMemcpyAsync(d_new_off.p, new_off.data(),
+                           (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice,
+                           stream));
+  HIP_CHECK(hipMemcpyAsync(d_old_off.p, old_off.data(),
+                           (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice,
+                           stream));
+  HIP_CHECK(hipMemcpyAsync(d_pend_off.p, pend_off.data(),
+                           (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice,
+                           stream));
+  if (n_pend)
+    HIP_CHECK(hipMemcpyAsync(d_psrc.p, psrc.data(), (size_t)n_pend * 4,
+                             hipMemcpyHostToDevice, stream));
+  h->cr_ids_new.ensure((size_t)std::max<int64_t>(n_new, 1) * 8);
+  h->id2pos.ensure((size_t)std::max<int64_t>(n_new, 1) * 4);
+  SlabArena next;
+  next.init(h->stride, h->csr_arena.slab_bytes());
+  const int64_t rps = next.rps();
+  int64_t id_base = n_old;  // pending row p arrived as id n_old + p
+  size_t old_free_cursor = 0;
+  const uint8_t *const *old_tab = h->csr_arena.dev_table(stream);
+  const uint8_t *const *pend_tab = h->pend_arena.dev_table(stream);
+  for (int64_t j0 = 0; j0 < n_new; j0 += rps) {
+    int64_t j1 = std::min(n_new, j0 + rps);
+    next.ensure_rows(j1);
+    hipLaunchKernelGGL(k_rebuild_gather, grid1d(j1 - j0), dim3(256), 0, stream,
+                       j0, j1, d_new_off.as<int64_t>(), d_old_off.as<int64_t>(),
+                       d_pend_off.as<int64_t>(), d_psrc.as<unsigned>(),
+                       old_tab, pend_tab, next.dev_table_mut(stream),
+                       h->cr_ids.as<int64_t>(), id_base,
+                       h->cr_ids_new.as<int64_t>(), h->id2pos.as<unsigned>(),
+                       h->nlist, next.rlog, h->stride);
+    HIP_CHECK(hipGetLastError());
+    // recycle old slabs fully behind the window: every old row s maps to
+    // new pos >= its old pos, so old pos < j1 - n_pend are consumed
+    if (h->csr_arena.rlog == next.rlog) {
+      int64_t consumed = j1 - n_pend;
+      bool freed_any = false;
+      while ((int64_t)((old_free_cursor + 1) << next.rlog) <= consumed &&
+             old_free_cursor < h->csr_arena.slabs.size()) {
+        if (!freed_any) {
+          HIP_CHECK(hipStreamSynchronize(stream));
+          freed_any = true;
+        }
+        h->csr_arena.free_slab(old_free_cursor++);
+      }
+    }
+  }
+  HIP_CHECK(hipStreamSynchronize(stream));
+  h->csr_arena.reset();
+  std::swap(h->csr_arena.stride, next.stride);
+  std::swap(h->csr_arena.rlog, next.rlog);
+  std::swap(h->csr_arena.slabs, next.slabs);
+  h->csr_arena.rows = n_new;
+  h->csr_arena.table_dirty = true;
+  next.rows = 0;
+  h->pend_arena.reset();
+  h->pend_arena.init(h->stride, h->csr_arena.slab_bytes());
+  h->h_pend_assign.clear();
+  h->csr_rows = n_new;
+  // swap id arrays (DevBuf members)
+  std::swap(h->cr_ids.p, h->cr_ids_new.p);
+  std::swap(h->cr_ids.cap, h->cr_ids_new.cap);
+  h->h_off = new_off;
+  h->cr_off.ensure((size_t)(h->nlist + 1) * 8);
+  HIP_CHECK(hipMemcpy(h->cr_off.p, h->h_off.data(), (size_t)(h->nlist + 1) * 8,
+                      hipMemcpyHostToDevice));
+  h->dirty = false;
+}
+
+// encode a chunk of rows into the pending arena at rows [row0, row0+n)
+static void encode_chunk(dfann_index *h, int64_t n, const float *x,
+                         const int *asg, int64_t row0, hipStream_t stream) {
+  h->pend_arena.ensure_rows(row0 + n);
+  uint8_t *const *dst = h->pend_arena.dev_table_mut(stream);
+  int rlog = h->pend_arena.rlog;
+  if (h->type == T_IVFFLAT) {
+    hipLaunchKernelGGL(k_pack_rows, grid1d(n * h->d), dim3(256), 0, stream, x,
+                       n, h->d, h->stride, rlog, row0, dst);
+    return;
+  }
+  DevBuf resid;
+  resid.ensure((size_t)n * h->d * 4);
+  hipLaunchKernelGGL(k_residual, grid1d(n * h->d), dim3(256), 0, stream, x,
+                     h->centroids.as<float>(), asg, n, h->d,
+                     resid.as<float>());
+  if (h->type == T_IVFPQ) {
+    // per-subspace encode as a distance GEMM + running argmin (~20x a
+    // naive per-thread 256-way argmin kernel)
+    DevBuf sub, best, bestv, cbn;
+    sub.ensure((size_t)n * h->dsub * 4);
+    best.ensure((size_t)n * 4);
+    bestv.ensure((size_t)n * 4);
+    cbn.ensure(256 * 4);
+    int64_t chunk = std::max<int64_t>(1, ((int64_t)h->ws_mb << 20) / (256 * 4));
+    chunk = std::min<int64_t>(chunk, n);
+    h->ws1.ensure((size_t)chunk * 256 * 4);
+    for (int j = 0; j < h->m; ++j) {
+      hipLaunchKernelGGL(k_subspace_slice, grid1d(n * h->dsub), dim3(256), 0,
+                         stream, resid.as<float>(), n, h->d, j * h->dsub,
+                         h->dsub, sub.as<float>());
+      const float *cbj = h->codebooks.as<float>() + (size_t)j * 256 * h->dsub;
+      rownorms(cbj, 256, h->dsub, cbn.as<float>(), stream);
+      hipLaunchKernelGGL(k_assign_init, grid1d(n), dim3(256), 0, stream,
+                         bestv.as<float>(), best.as<int>(), n);
+      for (int64_t s0 = 0; s0 < n; s0 += chunk) {
+        int64_t c = std::min(chunk, n - s0);
+        gemm_keys(nullptr, sub.as<float>() + s0 * h->dsub, c, cbj, 256,
+                  h->dsub, cbn.as<float>(), nullptr, 1, h->ws1.as<float>(),
+                  stream);
+        hipLaunchKernelGGL(k_assign_rowblock, dim3((unsigned)c), dim3(256), 0,
+                           stream, h->ws1.as<float>(), c, (long long)256,
+                           (long long)256, 0, bestv.as<float>() + s0,
+                           best.as<int>() + s0);
+      }
+      hipLaunchKernelGGL(k_codes_from_best, grid1d(n), dim3(256), 0, stream,
+                         best.as<int>(), n, j, h->stride, rlog, row0, dst);
+    }
+  } else {
+    hipLaunchKernelGGL(k_sq_encode, grid1d(n * h->d), dim3(256), 0, stream,
+                       resid.as<float>(),
+                       h->sq8 ? h->sq_vmin.as<float>() : nullptr,
+                       h->sq8 ? h->sq_vdiff.as<float>() : nullptr, n, h->d,
+                       h->stride, h->sq8 ? 0 : 1, rlog, row0, dst);
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
 static void add_impl(dfann_index *h, int64_t n, const float *x,
                      hipStream_t stream) {
   if (!h->trained) throw std::runtime_error("add on untrained index");
@@ -610,103 +880,35 @@ static void add_impl(dfann_index *h, int64_t n, const float *x,
     h->ntotal += n;
     return;
   }
+  // internal chunking: bounds the residual/assign transients for huge
+  // ingests (a 125M x 128 add would otherwise stage a 64 GB fp32
+  // residual buffer)
+  const int64_t CH = std::max<int64_t>(
+      65536, ((int64_t)h->ws_mb << 20) / std::max(4 * h->d, 1));
   DevBuf asg;
-  asg.ensure((size_t)n * 4);
-  assign_rows(h, x, n, asg.as<int>(), stream);
-  h->st_codes.grow_keep(((size_t)h->ntotal + n) * h->stride,
-                        (size_t)h->ntotal * h->stride);
-  uint8_t *dst = h->st_codes.as<uint8_t>() + (size_t)h->ntotal * h->stride;
-  if (h->type == T_IVFFLAT) {
-    hipLaunchKernelGGL(k_pack_rows, grid1d(n * h->d), dim3(256), 0, stream, x,
-                       n, h->d, h->stride, dst);
-  } else {
-    DevBuf resid;
-    resid.ensure((size_t)n * h->d * 4);
-    hipLaunchKernelGGL(k_residual, grid1d(n * h->d), dim3(256), 0, stream, x,
-                       h->centroids.as<float>(), asg.as<int>(), n, h->d,
-                       resid.as<float>());
-    if (h->type == T_IVFPQ) {
-      // per-subspace encode as a distance GEMM + running argmin (~20x a
-      // naive per-thread 256-way argmin kernel)
-      DevBuf sub, best, bestv, cbn;
-      sub.ensure((size_t)n * h->dsub * 4);
-      best.ensure((size_t)n * 4);
-      bestv.ensure((size_t)n * 4);
-      cbn.ensure(256 * 4);
-      int64_t chunk = std::max<int64_t>(
-          1, ((int64_t)h->ws_mb << 20) / (256 * 4));
-      chunk = std::min<int64_t>(chunk, n);
-      h->ws1.ensure((size_t)chunk * 256 * 4);
-      for (int j = 0; j < h->m; ++j) {
-        hipLaunchKernelGGL(k_subspace_slice, grid1d(n * h->dsub), dim3(256), 0,
-                           stream, resid.as<float>(), n, h->d, j * h->dsub,
-                           h->dsub, sub.as<float>());
-        const float *cbj = h->codebooks.as<float>() + (size_t)j * 256 * h->dsub;
-        rownorms(cbj, 256, h->dsub, cbn.as<float>(), stream);
-        hipLaunchKernelGGL(k_assign_init, grid1d(n), dim3(256), 0, stream,
-                           bestv.as<float>(), best.as<int>(), n);
-        for (int64_t s0 = 0; s0 < n; s0 += chunk) {
-          int64_t c = std::min(chunk, n - s0);
-          gemm_keys(nullptr, sub.as<float>() + s0 * h->dsub, c, cbj, 256,
-                    h->dsub, cbn.as<float>(), nullptr, 1, h->ws1.as<float>(),
-                    stream);
-          hipLaunchKernelGGL(k_assign_rowblock, dim3((unsigned)c), dim3(256),
-                             0, stream, h->ws1.as<float>(), c, (long long)256,
-                             (long long)256, 0, bestv.as<float>() + s0,
-                             best.as<int>() + s0);
-        }
-        hipLaunchKernelGGL(k_codes_from_best, grid1d(n), dim3(256), 0, stream,
-                           best.as<int>(), n, j, h->stride, dst);
-      }
-      HIP_CHECK(hipGetLastError());
-    } else {
-      hipLaunchKernelGGL(k_sq_encode, grid1d(n * h->d), dim3(256), 0, stream,
-                         resid.as<float>(),
-                         h->sq8 ? h->sq_vmin.as<float>() : nullptr,
-                         h->sq8 ? h->sq_vdiff.as<float>() : nullptr, n, h->d,
-                         h->stride, h->sq8 ? 0 : 1, dst);
-    }
+  for (int64_t s = 0; s < n; s += CH) {
+    int64_t c = std::min(CH, n - s);
+    asg.ensure((size_t)c * 4);
+    assign_rows(h, x + s * h->d, c, asg.as<int>(), stream);
+    int64_t row0 = (int64_t)h->h_pend_assign.size();
+    encode_chunk(h, c, x + s * h->d, asg.as<int>(), row0, stream);
+    size_t old = h->h_pend_assign.size();
+    h->h_pend_assign.resize(old + c);
+    HIP_CHECK(hipMemcpyAsync(h->h_pend_assign.data() + old, asg.p,
+                             (size_t)c * 4, hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    h->ntotal += c;
+    h->dirty = true;
+    // incremental merge: bounds the pending arena (and so the rebuild
+    // transient) to ~merge_mb
+    if ((int64_t)h->h_pend_assign.size() * h->stride >=
+        ((int64_t)h->merge_mb << 20))
+      rebuild_csr(h, stream);
   }
-  HIP_CHECK(hipGetLastError());
-  // download assignments (arrival-order host copy drives finalize)
-  size_t old = h->h_assign.size();
-  h->h_assign.resize(old + n);
-  HIP_CHECK(hipMemcpyAsync(h->h_assign.data() + old, asg.p, (size_t)n * 4,
-                           hipMemcpyDeviceToHost, stream));
-  HIP_CHECK(hipStreamSynchronize(stream));
-  h->ntotal += n;
-  h->dirty = true;
 }
 
-// rebuild CSR from staging (stable counting sort by list)
 static void finalize_csr(dfann_index *h, hipStream_t stream) {
-  if (!h->dirty) return;
-  int64_t n = h->ntotal;
-  h->h_off.assign(h->nlist + 1, 0);
-  for (int64_t i = 0; i < n; ++i) h->h_off[h->h_assign[i] + 1]++;
-  for (int l = 0; l < h->nlist; ++l) h->h_off[l + 1] += h->h_off[l];
-  std::vector<unsigned> src(n);
-  {
-    std::vector<int64_t> fill(h->h_off.begin(), h->h_off.end() - 1);
-    for (int64_t i = 0; i < n; ++i) src[fill[h->h_assign[i]]++] = (unsigned)i;
-  }
-  h->cr_codes.ensure((size_t)n * h->stride);
-  h->cr_ids.ensure((size_t)n * 8);
-  h->id2pos.ensure((size_t)n * 4);
-  h->cr_off.ensure((size_t)(h->nlist + 1) * 8);
-  DevBuf src_dev;
-  src_dev.ensure((size_t)n * 4);
-  HIP_CHECK(hipMemcpyAsync(src_dev.p, src.data(), (size_t)n * 4,
-                           hipMemcpyHostToDevice, stream));
-  HIP_CHECK(hipMemcpyAsync(h->cr_off.p, h->h_off.data(),
-                           (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice,
-                           stream));
-  hipLaunchKernelGGL(k_gather_finalize, grid1d(n), dim3(256), 0, stream,
-                     h->st_codes.as<uint8_t>(), src_dev.as<unsigned>(), n,
-                     h->stride, h->cr_codes.as<uint8_t>(),
-                     h->cr_ids.as<int64_t>(), h->id2pos.as<unsigned>());
-  HIP_CHECK(hipStreamSynchronize(stream));
-  h->dirty = false;
+  rebuild_csr(h, stream);
 }
 
 // ---------------------------------------------------------------------------
@@ -767,9 +969,9 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   unsigned *cand_p;
   int fam_floats;
   void (*kern)(const float *, const float *, const float *, const float *,
-               const float *, const int *, const float *, const uint8_t *,
-               const int64_t *, int, int, int, int, int, int, int, float *,
-               unsigned *, int, int) = nullptr;
+               const float *, const int *, const float *,
+               const uint8_t *const *, const int64_t *, int, int, int, int,
+               int, int, int, int, float *, unsigned *, int, int) = nullptr;
   bool ip = h->metric == M_IP;
   bool rk = use_regsel(k);
   switch (h->type) {
@@ -866,8 +1068,10 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        stream, q, h->centroids.as<float>(),
                        h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                        h->sq_scale.as<float>(), probes, keys,
-                       h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
+                       h->csr_arena.dev_table(stream),
+                       h->cr_off.as<int64_t>(),
                        (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
+                       h->csr_arena.rlog,
                        cand_d, cand_p, fam_floats, h->term2.as<float>(),
                        h->term3_ws.as<float>(), h->qn_ws.as<float>());
     if (h->timing) h->ev_end(e, stream, h->ev_scan);
@@ -915,19 +1119,21 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
       if (h->timing) e = h->ev_begin(stream);
       // persistent multi-pair variant (fp16 LUT + register top-k): a
       // fixed grid strides over the (query, probe) pairs and prefetches
-      // the next pair's LUT during the current scan — the one-pair
-      // kernel is staging-dominated at short lists (see kernels.hip).
-      // Bit-identical results; DFANN_SCAN_PERS=0 forces the old path.
+      // the next pair's LUT during the current scan. MEASURED NEGATIVE
+      // at the headline shape (gpurun_out/r2s_pers*.json: scan 5.7 ->
+      // 6.9 ms at 2048 blocks — the hardware's cross-block overlap of
+      // 80k one-pair blocks beats the in-block pipeline), so OFF by
+      // default; DFANN_SCAN_PERS=1 re-enables for experiments.
+      // Bit-identical results either way (r2s_pers_check.log).
       int pfn = 0;
       long long qpn_c = (long long)nqc * nprobe;
-      if (lut_f16 && rk) {
+      const char *pers_env = getenv("DFANN_SCAN_PERS");
+      if (lut_f16 && rk && pers_env && atoi(pers_env) == 1) {
         int lut_u4 = h->m * 32;  // m*256 halves / 8 per uint4
         if (lut_u4 % (int)scan_bs == 0) {
           int cand = lut_u4 / (int)scan_bs;
           if (cand == 2 || cand == 4 || cand == 8 || cand == 16) pfn = cand;
         }
-        if (const char *e2 = getenv("DFANN_SCAN_PERS"))
-          if (atoi(e2) == 0) pfn = 0;
       }
       if (pfn) {
         auto pk = ip ? (pfn == 2 ? k_scan_pq_ip_ghp2
@@ -944,9 +1150,9 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                            q + q0 * h->d, h->centroids.as<float>(),
                            h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                            h->sq_scale.as<float>(), probes + q0 * nprobe,
-                           keys + q0 * nprobe, h->cr_codes.as<uint8_t>(),
+                           keys + q0 * nprobe, h->csr_arena.dev_table(stream),
                            h->cr_off.as<int64_t>(), (int)nqc, nprobe, h->d,
-                           h->m, h->dsub, k, h->stride,
+                           h->m, h->dsub, k, h->stride, h->csr_arena.rlog,
                            cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
                            fam_floats, lutg);
       } else {
@@ -954,9 +1160,9 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                            lds, stream, q + q0 * h->d, h->centroids.as<float>(),
                            h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                            h->sq_scale.as<float>(), probes + q0 * nprobe,
-                           keys + q0 * nprobe, h->cr_codes.as<uint8_t>(),
+                           keys + q0 * nprobe, h->csr_arena.dev_table(stream),
                            h->cr_off.as<int64_t>(), (int)nqc, nprobe, h->d,
-                           h->m, h->dsub, k, h->stride,
+                           h->m, h->dsub, k, h->stride, h->csr_arena.rlog,
                            cand_d + q0 * nprobe * k, cand_p + q0 * nprobe * k,
                            fam_floats, lutg);
       }
@@ -968,9 +1174,9 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
                        dim3(scan_bs), lds, stream, q, h->centroids.as<float>(),
                        h->codebooks.as<float>(), h->sq_vmin.as<float>(),
                        h->sq_scale.as<float>(), probes, keys,
-                       h->cr_codes.as<uint8_t>(), h->cr_off.as<int64_t>(),
+                       h->csr_arena.dev_table(stream), h->cr_off.as<int64_t>(),
                        (int)nq, nprobe, h->d, h->m, h->dsub, k, h->stride,
-                       cand_d, cand_p, fam_floats, fan);
+                       h->csr_arena.rlog, cand_d, cand_p, fam_floats, fan);
     if (h->timing) h->ev_end(e, stream, h->ev_scan);
   }
   if (h->timing) {
@@ -1177,7 +1383,8 @@ extern "C" int dfann_search_reconstruct(dfann_index *h, int64_t nq,
   else rtype = h->sq8 ? 3 : 4;
   hipLaunchKernelGGL(k_reconstruct, dim3((unsigned)(nq * k)), dim3(64), 0,
                      (hipStream_t)stream, I_dev, nq, k, rtype, h->d, h->m,
-                     h->dsub, h->stride, flat_src, h->cr_codes.as<uint8_t>(),
+                     h->dsub, h->stride, h->csr_arena.rlog, flat_src,
+                     h->csr_arena.dev_table((hipStream_t)stream),
                      h->id2pos.as<unsigned>(), h->cr_off.as<int64_t>(),
                      h->nlist, h->centroids.as<float>(),
                      h->codebooks.as<float>(), h->sq_vmin.as<float>(),
@@ -1227,8 +1434,7 @@ extern "C" int dfann_get_lists(dfann_index *h, int64_t *off_host,
   if (h->ntotal) {
     HIP_CHECK(hipMemcpy(ids_host, h->cr_ids.p, (size_t)h->ntotal * 8,
                         hipMemcpyDeviceToHost));
-    HIP_CHECK(hipMemcpy(codes_host, h->cr_codes.p,
-                        (size_t)h->ntotal * h->stride, hipMemcpyDeviceToHost));
+    h->csr_arena.copy_rows_to_host(codes_host, 0, h->ntotal);
   }
   API_END
 }
@@ -1366,7 +1572,15 @@ extern "C" int dfann_save(dfann_index *h, const char *path) {
     } else if (h->ntotal) {
       fwrite_chk(h->h_off.data(), (size_t)(h->nlist + 1) * 8, f);
       dump_dev(f, h->cr_ids, (size_t)h->ntotal * 8);
-      dump_dev(f, h->cr_codes, (size_t)h->ntotal * h->stride);
+      // codes: slab arena -> file in CSR row order (same byte layout as
+      // the round-1 contiguous image)
+      const int64_t CHROWS = std::max<int64_t>(1, ((int64_t)64 << 20) / h->stride);
+      std::vector<char> tmp((size_t)CHROWS * h->stride);
+      for (int64_t r = 0; r < h->ntotal; r += CHROWS) {
+        int64_t c = std::min(CHROWS, h->ntotal - r);
+        h->csr_arena.copy_rows_to_host(tmp.data(), r, c);
+        fwrite_chk(tmp.data(), (size_t)c * h->stride, f);
+      }
     }
   } catch (...) {
     fclose(f);
@@ -1423,26 +1637,30 @@ extern "C" int dfann_load(const char *path, dfann_index **out) {
       h->h_off.resize(h->nlist + 1);
       fread_chk(h->h_off.data(), (size_t)(h->nlist + 1) * 8, f);
       load_dev(f, h->cr_ids, (size_t)h->ntotal * 8);
-      load_dev(f, h->cr_codes, (size_t)h->ntotal * h->stride);
+      {
+        // codes: file (CSR row order) -> slab arena
+        const int64_t CHROWS =
+            std::max<int64_t>(1, ((int64_t)64 << 20) / h->stride);
+        std::vector<char> tmp((size_t)CHROWS * h->stride);
+        for (int64_t r = 0; r < h->ntotal; r += CHROWS) {
+          int64_t c = std::min(CHROWS, h->ntotal - r);
+          fread_chk(tmp.data(), (size_t)c * h->stride, f);
+          h->csr_arena.copy_rows_from_host(tmp.data(), r, c);
+        }
+        h->csr_arena.rows = h->ntotal;
+        h->csr_rows = h->ntotal;
+      }
       h->cr_off.ensure((size_t)(h->nlist + 1) * 8);
       HIP_CHECK(hipMemcpy(h->cr_off.p, h->h_off.data(),
                           (size_t)(h->nlist + 1) * 8, hipMemcpyHostToDevice));
       h->id2pos.ensure((size_t)h->ntotal * 4);
-      // rebuild arrival-order staging + h_assign from CSR
-      h->st_codes.ensure((size_t)h->ntotal * h->stride);
-      hipLaunchKernelGGL(k_scatter_rows, grid1d(h->ntotal), dim3(256), 0, 0,
-                         h->cr_codes.as<uint8_t>(), h->cr_ids.as<int64_t>(),
-                         h->ntotal, h->stride, h->st_codes.as<uint8_t>());
       std::vector<int64_t> ids(h->ntotal);
       HIP_CHECK(hipMemcpy(ids.data(), h->cr_ids.p, (size_t)h->ntotal * 8,
                           hipMemcpyDeviceToHost));
-      h->h_assign.resize(h->ntotal);
       std::vector<unsigned> i2p(h->ntotal);
       for (int l = 0; l < h->nlist; ++l)
-        for (int64_t j = h->h_off[l]; j < h->h_off[l + 1]; ++j) {
-          h->h_assign[ids[j]] = l;
+        for (int64_t j = h->h_off[l]; j < h->h_off[l + 1]; ++j)
           i2p[ids[j]] = (unsigned)j;
-        }
       HIP_CHECK(hipMemcpy(h->id2pos.p, i2p.data(), (size_t)h->ntotal * 4,
                           hipMemcpyHostToDevice));
       HIP_CHECK(hipDeviceSynchronize());

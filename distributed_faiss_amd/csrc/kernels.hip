@@ -25,6 +25,28 @@
 #define PAD_POS 0xFFFFFFFFu
 #define DFANN_FLT_MAX 3.402823466e+38f
 
+// ---------------------------------------------------------------------------
+// slab-arena addressing: code images live in fixed-size slabs of
+// (1 << rlog) rows each (DESIGN.md §2 memory plan: the arena grows by
+// whole slabs and is recycled slab-by-slab during CSR rebuilds, so the
+// engine never holds a 2x contiguous copy). `pos` stays the global CSR
+// row index; kernels translate through the device slab-pointer table
+// (a few hundred entries, L1-resident).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ const uint8_t *slab_row(
+    const uint8_t *const *__restrict__ slabs, int rlog, long long pos,
+    int stride) {
+  return slabs[pos >> rlog] +
+         (size_t)(pos & ((1LL << rlog) - 1)) * (size_t)stride;
+}
+
+__device__ __forceinline__ uint8_t *slab_row_mut(
+    uint8_t *const *__restrict__ slabs, int rlog, long long pos, int stride) {
+  return slabs[pos >> rlog] +
+         (size_t)(pos & ((1LL << rlog) - 1)) * (size_t)stride;
+}
+
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 // ---------------------------------------------------------------------------
@@ -1001,9 +1023,9 @@ __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
     const float *__restrict__ sq_scale, const int *__restrict__ probes,
-    const float *__restrict__ keys, const uint8_t *__restrict__ codes,
+    const float *__restrict__ keys, const uint8_t *const *__restrict__ codes,
     const int64_t *__restrict__ off, int nq, int nprobe, int d, int m,
-    int dsub, int k, int stride, float *__restrict__ cand_d,
+    int dsub, int k, int stride, int rlog, float *__restrict__ cand_d,
     unsigned *__restrict__ cand_p, int fam_floats,
     const float *__restrict__ term2 = nullptr,
     const float *__restrict__ term3 = nullptr,
@@ -1175,7 +1197,7 @@ __device__ void ivf_scan_body(
         float dist = 0.f;
         bool valid = pos < s1;
         if (valid) {
-          const float *vp = reinterpret_cast<const float *>(codes + pos * (size_t)stride);
+          const float *vp = reinterpret_cast<const float *>(slab_row(codes, rlog, pos, stride));
           float part = 0.f;
           for (int t = sub; t < d; t += 16) {
 #pragma clang fp contract(off)
@@ -1220,7 +1242,7 @@ __device__ void ivf_scan_body(
             int t0 = g8 * 16;
             wv4[u] = (val4[u] && t0 < d)
                          ? *reinterpret_cast<const uint4 *>(
-                               codes + pos * (size_t)stride + t0)
+                               slab_row(codes, rlog, pos, stride) + t0)
                          : uint4{0, 0, 0, 0};
           }
 #pragma unroll
@@ -1263,7 +1285,7 @@ __device__ void ivf_scan_body(
             bool valid = pos < s1;
             float part = 0.f;
             if (valid) {
-              const uint8_t *cp = codes + pos * (size_t)stride;
+              const uint8_t *cp = slab_row(codes, rlog, pos, stride);
               for (int t0 = g8 * 16; t0 < d; t0 += 128) {
 #pragma clang fp contract(off)
                 uint4 wv = *reinterpret_cast<const uint4 *>(cp + t0);
@@ -1310,7 +1332,7 @@ __device__ void ivf_scan_body(
         if (u >= UROWS) break;
         long long pos = base + (long long)u * BS + threadIdx.x;
         if (pos < s1) {
-          const uint8_t *cp = codes + pos * (size_t)stride;
+          const uint8_t *cp = slab_row(codes, rlog, pos, stride);
           float acc = scan_row_dist<FAM, IS_IP, L16>(cp, fam, d, m);
           float dist = IS_IP ? -(bias + acc) : (PRE ? bias + acc : acc);
           loc.push(dist, (unsigned)pos);
@@ -1325,7 +1347,7 @@ __device__ void ivf_scan_body(
       for (int u = 0; u < 2; ++u) {
         long long pos = base + (long long)u * BS + threadIdx.x;
         if (pos < s1) {
-          const uint8_t *cp = codes + pos * (size_t)stride;
+          const uint8_t *cp = slab_row(codes, rlog, pos, stride);
           float acc = scan_row_dist<FAM, IS_IP, L16>(cp, fam, d, m);
           float dist = IS_IP ? -(bias + acc) : (PRE ? bias + acc : acc);
           sel_try(s, dist, (unsigned)pos);
@@ -1352,12 +1374,13 @@ __device__ void ivf_scan_body(
   extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
-      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
-      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
-      unsigned *cand_p, int fam_floats, int fan) {                             \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,          \
+      float *cand_d, unsigned *cand_p, int fam_floats, int fan) {                             \
     ivf_scan_body<FAM, IS_IP, REGSEL>(q, cent, cb, sq_vmin, sq_scale, probes,  \
                                       keys, codes, off, nq, nprobe, d, m,      \
-                                      dsub, k, stride, cand_d, cand_p,         \
+                                      dsub, k, stride, rlog, cand_d, cand_p,   \
                                       fam_floats, nullptr, nullptr, nullptr,   \
                                       fan);                                    \
   }
@@ -1366,13 +1389,14 @@ __device__ void ivf_scan_body(
   extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
-      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
-      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
-      unsigned *cand_p, int fam_floats, const float *term2,                    \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,          \
+      float *cand_d, unsigned *cand_p, int fam_floats, const float *term2,     \
       const float *term3, const float *qn) {                                   \
     ivf_scan_body<0, false, REGSEL, true>(q, cent, cb, sq_vmin, sq_scale,      \
                                           probes, keys, codes, off, nq,        \
-                                          nprobe, d, m, dsub, k, stride,       \
+                                          nprobe, d, m, dsub, k, stride, rlog, \
                                           cand_d, cand_p, fam_floats, term2,   \
                                           term3, qn);                          \
   }
@@ -1398,13 +1422,14 @@ INSTANTIATE_SCAN(k_scan_pq_ip_rk, 0, true, true)
   extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
-      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
-      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
-      unsigned *cand_p, int fam_floats, const float *glut) {                   \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,          \
+      float *cand_d, unsigned *cand_p, int fam_floats, const float *glut) {                   \
     ivf_scan_body<0, IS_IP, REGSEL, false, true>(                              \
         q, cent, cb, sq_vmin, sq_scale, probes, keys, codes, off, nq, nprobe,  \
-        d, m, dsub, k, stride, cand_d, cand_p, fam_floats, nullptr, nullptr,   \
-        nullptr, 1, glut);                                                     \
+        d, m, dsub, k, stride, rlog, cand_d, cand_p, fam_floats, nullptr,      \
+        nullptr, nullptr, 1, glut);                                                     \
   }
 INSTANTIATE_SCAN_GLUT(k_scan_pq_l2_g, false, false)
 INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g, true, false)
@@ -1416,13 +1441,14 @@ INSTANTIATE_SCAN_GLUT(k_scan_pq_ip_g_rk, true, true)
   extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
-      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
-      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
-      unsigned *cand_p, int fam_floats, const float *glut) {                   \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,          \
+      float *cand_d, unsigned *cand_p, int fam_floats, const float *glut) {                   \
     ivf_scan_body<0, IS_IP, REGSEL, false, true, true>(                        \
         q, cent, cb, sq_vmin, sq_scale, probes, keys, codes, off, nq, nprobe,  \
-        d, m, dsub, k, stride, cand_d, cand_p, fam_floats, nullptr, nullptr,   \
-        nullptr, 1, glut);                                                     \
+        d, m, dsub, k, stride, rlog, cand_d, cand_p, fam_floats, nullptr,      \
+        nullptr, nullptr, 1, glut);                                                     \
   }
 INSTANTIATE_SCAN_GLUT_H(k_scan_pq_l2_gh, false, false)
 INSTANTIATE_SCAN_GLUT_H(k_scan_pq_ip_gh, true, false)
@@ -1454,9 +1480,9 @@ __device__ void scan_pq_ghp_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
     const float *__restrict__ sq_scale, const int *__restrict__ probes,
-    const float *__restrict__ keys, const uint8_t *__restrict__ codes,
+    const float *__restrict__ keys, const uint8_t *const *__restrict__ codes,
     const int64_t *__restrict__ off, int nq, int nprobe, int d, int m,
-    int dsub, int k, int stride, float *__restrict__ cand_d,
+    int dsub, int k, int stride, int rlog, float *__restrict__ cand_d,
     unsigned *__restrict__ cand_p, int fam_floats,
     const float *__restrict__ glut) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -1508,7 +1534,7 @@ __device__ void scan_pq_ghp_body(
         for (int u = 0; u < 4; ++u) {
           long long pos = base + (long long)u * BS + threadIdx.x;
           if (pos < s1) {
-            const uint8_t *cp = codes + pos * (size_t)stride;
+            const uint8_t *cp = slab_row(codes, rlog, pos, stride);
             float acc = scan_row_dist<0, IS_IP, true>(cp, fam, d, m);
             float dist = IS_IP ? -(bias + acc) : acc;
             loc.push(dist, (unsigned)pos);
@@ -1533,12 +1559,14 @@ __device__ void scan_pq_ghp_body(
   extern "C" __global__ __launch_bounds__(512) void NAME(                      \
       const float *q, const float *cent, const float *cb,                      \
       const float *sq_vmin, const float *sq_scale, const int *probes,          \
-      const float *keys, const uint8_t *codes, const int64_t *off, int nq,     \
-      int nprobe, int d, int m, int dsub, int k, int stride, float *cand_d,    \
-      unsigned *cand_p, int fam_floats, const float *glut) {                   \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,     \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,          \
+      float *cand_d, unsigned *cand_p, int fam_floats, const float *glut) {                   \
     scan_pq_ghp_body<IS_IP, PFN>(q, cent, cb, sq_vmin, sq_scale, probes,       \
                                  keys, codes, off, nq, nprobe, d, m, dsub, k,  \
-                                 stride, cand_d, cand_p, fam_floats, glut);    \
+                                 stride, rlog, cand_d, cand_p, fam_floats,     \
+                                 glut);    \
   }
 
 INSTANTIATE_SCAN_GHP(k_scan_pq_l2_ghp2, false, 2)
@@ -1905,13 +1933,15 @@ extern "C" __global__ void k_residual(const float *__restrict__ x,
 }
 
 // write subspace argmin results into the packed code column j (PQ
-// encode = per-subspace distance GEMM + k_assign_rowblock + this)
+// encode = per-subspace distance GEMM + k_assign_rowblock + this);
+// writes rows [row0, row0+n) of the slab arena
 extern "C" __global__ void k_codes_from_best(const int *__restrict__ best,
                                              long long n, int j, int stride,
-                                             uint8_t *__restrict__ codes) {
+                                             int rlog, long long row0,
+                                             uint8_t *const *__restrict__ codes) {
   long long p = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   for (; p < n; p += (long long)gridDim.x * blockDim.x)
-    codes[p * (size_t)stride + j] = (uint8_t)best[p];
+    slab_row_mut(codes, rlog, row0 + p, stride)[j] = (uint8_t)best[p];
 }
 
 // SQ encode (8bit / fp16) of residuals
@@ -1919,22 +1949,23 @@ extern "C" __global__ void k_sq_encode(const float *__restrict__ resid,
                                        const float *__restrict__ vmin,
                                        const float *__restrict__ vdiff,
                                        long long n, int d, int stride,
-                                       int is_fp16, uint8_t *__restrict__ codes) {
+                                       int is_fp16, int rlog, long long row0,
+                                       uint8_t *const *__restrict__ codes) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   long long total = n * d;
   for (; i < total; i += (long long)gridDim.x * blockDim.x) {
     long long r = i / d;
     int t = (int)(i % d);
     float v = resid[i];
+    uint8_t *row = slab_row_mut(codes, rlog, row0 + r, stride);
     if (is_fp16) {
       __half h = __float2half(v);
-      reinterpret_cast<unsigned short *>(codes + r * (size_t)stride)[t] =
-          __half_as_ushort(h);
+      reinterpret_cast<unsigned short *>(row)[t] = __half_as_ushort(h);
     } else {
       float xi = (v - vmin[t]) / vdiff[t];
       int c = (int)(255.0f * xi);  // trunc toward zero, as the oracle
       c = c < 0 ? 0 : (c > 255 ? 255 : c);
-      codes[r * (size_t)stride + t] = (uint8_t)c;
+      row[t] = (uint8_t)c;
     }
   }
 }
@@ -1985,46 +2016,66 @@ extern "C" __global__ void k_pq_term3(const float *__restrict__ q,
   }
 }
 
-// pack raw fp32 rows into the staging byte arena (IVF-Flat codes)
+// pack raw fp32 rows into the pending byte arena (IVF-Flat codes)
 extern "C" __global__ void k_pack_rows(const float *__restrict__ x, long long n,
-                                       int d, int stride,
-                                       uint8_t *__restrict__ out) {
+                                       int d, int stride, int rlog,
+                                       long long row0,
+                                       uint8_t *const *__restrict__ out) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   long long total = n * d;
   for (; i < total; i += (long long)gridDim.x * blockDim.x) {
     long long r = i / d;
     int t = (int)(i % d);
-    reinterpret_cast<float *>(out + r * (size_t)stride)[t] = x[i];
+    reinterpret_cast<float *>(slab_row_mut(out, rlog, row0 + r, stride))[t] =
+        x[i];
   }
 }
 
-// finalize gather: CSR[j] = staging[src[j]]; ids[j] = src[j]; id2pos[src[j]] = j
-extern "C" __global__ void k_gather_finalize(const uint8_t *__restrict__ staging,
-                                             const unsigned *__restrict__ src,
-                                             long long n, int stride,
-                                             uint8_t *__restrict__ csr,
-                                             int64_t *__restrict__ ids,
-                                             unsigned *__restrict__ id2pos) {
-  long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= n) return;
-  unsigned sidx = src[j];
-  ids[j] = (long long)sidx;
-  id2pos[sidx] = (unsigned)j;
-  const uint4 *in = reinterpret_cast<const uint4 *>(staging + (size_t)sidx * stride);
-  uint4 *out = reinterpret_cast<uint4 *>(csr + (size_t)j * stride);
-  for (int t = 0; t < stride / 16; ++t) out[t] = in[t];
-}
-
-// inverse of finalize: rebuild the arrival-order staging arena from CSR
-extern "C" __global__ void k_scatter_rows(const uint8_t *__restrict__ csr,
-                                          const int64_t *__restrict__ ids,
-                                          long long n, int stride,
-                                          uint8_t *__restrict__ staging) {
-  long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= n) return;
-  const uint4 *in = reinterpret_cast<const uint4 *>(csr + (size_t)j * stride);
-  uint4 *out = reinterpret_cast<uint4 *>(staging + (size_t)ids[j] * stride);
-  for (int t = 0; t < stride / 16; ++t) out[t] = in[t];
+// two-source CSR rebuild gather (DESIGN.md §2 memory plan): new CSR row j
+// of list l is either an OLD CSR row (same list order) or a PENDING row
+// (arrival order within the list, via the psrc permutation). Both source
+// arenas and the destination are slab-based; the host launches one call
+// per new slab in ascending pos order and frees consumed old slabs
+// behind the window. ids: old rows keep their ids; pending row p gets
+// id_base + p (ids ARE arrival positions). id2pos rewritten for all.
+extern "C" __global__ void k_rebuild_gather(
+    long long j0, long long j1, const int64_t *__restrict__ new_off,
+    const int64_t *__restrict__ old_off, const int64_t *__restrict__ pend_off,
+    const unsigned *__restrict__ psrc,
+    const uint8_t *const *__restrict__ old_slabs,
+    const uint8_t *const *__restrict__ pend_slabs,
+    uint8_t *const *__restrict__ new_slabs,
+    const int64_t *__restrict__ old_ids, long long id_base,
+    int64_t *__restrict__ new_ids, unsigned *__restrict__ id2pos, int nlist,
+    int rlog, int stride) {
+  long long j = j0 + (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= j1) return;
+  // binary search: largest l with new_off[l] <= j
+  int lo = 0, hi = nlist;
+  while (lo + 1 < hi) {
+    int mid = (lo + hi) >> 1;
+    if (new_off[mid] <= j) lo = mid;
+    else hi = mid;
+  }
+  long long r = j - new_off[lo];
+  long long old_len = old_off[lo + 1] - old_off[lo];
+  const uint8_t *sp;
+  long long id;
+  if (r < old_len) {
+    long long src = old_off[lo] + r;
+    sp = slab_row(old_slabs, rlog, src, stride);
+    id = old_ids[src];
+  } else {
+    long long pi = pend_off[lo] + (r - old_len);
+    unsigned ps = psrc[pi];
+    sp = slab_row(pend_slabs, rlog, (long long)ps, stride);
+    id = id_base + (long long)ps;
+  }
+  uint8_t *dp = slab_row_mut(new_slabs, rlog, j, stride);
+  for (int t = 0; t < stride / 16; ++t)
+    reinterpret_cast<uint4 *>(dp)[t] = reinterpret_cast<const uint4 *>(sp)[t];
+  new_ids[j] = id;
+  id2pos[id] = (unsigned)j;
 }
 
 // gather rows by index (f32): out[i] = in[idx[i]]
@@ -2149,8 +2200,9 @@ extern "C" __global__ void k_minmax_decode(const unsigned *mn, const unsigned *m
 
 extern "C" __global__ __launch_bounds__(64) void k_reconstruct(
     const int64_t *__restrict__ I, long long nq, int k, int type, int d,
-    int m, int dsub, int stride, const float *__restrict__ flat_src,
-    const uint8_t *__restrict__ codes, const unsigned *__restrict__ id2pos,
+    int m, int dsub, int stride, int rlog, const float *__restrict__ flat_src,
+    const uint8_t *const *__restrict__ codes,
+    const unsigned *__restrict__ id2pos,
     const int64_t *__restrict__ off, int nlist,
     const float *__restrict__ cent, const float *__restrict__ cb,
     const float *__restrict__ vmin, const float *__restrict__ scale,
@@ -2177,7 +2229,7 @@ extern "C" __global__ __launch_bounds__(64) void k_reconstruct(
     else hi = mid;
   }
   int L = lo;
-  const uint8_t *cp = codes + (size_t)pos * stride;
+  const uint8_t *cp = slab_row(codes, rlog, (long long)pos, stride);
   if (type == 0) {  // ivfflat raw
     const float *src = reinterpret_cast<const float *>(cp);
     for (int t = threadIdx.x; t < d; t += blockDim.x) out[t] = src[t];

@@ -99,3 +99,43 @@ def test_checksum_of_checksums_full_probe():
     torch.cuda.synchronize()
     agree = (Ii.cpu().numpy() == If.cpu().numpy()).mean()
     assert agree > 0.999, f"exhaustive IVF vs flat agreement {agree}"
+
+
+def test_incremental_merge_equals_bulk():
+    # slab-arena memory plan: a tiny merge_mb forces many
+    # incremental two-source CSR rebuilds during interleaved adds; the
+    # final index must return BITWISE the same results as a bulk-added
+    # engine with the default merge cadence, across search + persistence
+    import numpy as np
+
+    from distributed_faiss_amd.hip_engine import HipEngine
+
+    rng = np.random.default_rng(5)
+    d, n, nq, k = 64, 120_000, 500, 10
+    xb = rng.standard_normal((n, d), dtype=np.float32)
+    q = rng.standard_normal((nq, d), dtype=np.float32)
+    for typ, extra in (("ivfpq", {"m": 8}), ("ivfsq", {"sq_type": "8bit"}),
+                       ("ivf_flat", {})):
+        spec = {"type": typ, "dim": d, "metric": 1, "nlist": 64, "nbits": 8,
+                "nprobe": 8, "seed": 3, **extra}
+        a = HipEngine(spec=dict(spec, merge_mb=1))  # 1 MB: merges often
+        b = HipEngine(spec=spec)
+        a.train(xb[:40_000])
+        cent = a.get_centroids()
+        cb = a.get_codebooks() if typ == "ivfpq" else None
+        vmin = vdiff = None
+        if typ == "ivfsq":
+            vmin, vdiff = a.get_sq_params()
+        b.set_trained(cent, cb, vmin, vdiff)
+        # a: many small adds with a search interleaved (forces rebuild
+        # with non-empty old CSR); b: one bulk add
+        CH = 7_000
+        for s in range(0, n, CH):
+            a.add(xb[s:s + CH])
+            if s == 3 * CH:
+                a.search(q[:8], k)
+        b.add(xb)
+        Da, Ia = a.search(q, k)
+        Db, Ib = b.search(q, k)
+        np.testing.assert_array_equal(Ia, Ib)
+        np.testing.assert_array_equal(Da, Db)
