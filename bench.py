@@ -92,6 +92,15 @@ WORKLOADS = {
         metric=1, nq=10_000, k=10, centers=12_500, sigma=0.5, latent=32,
         coarse_bf16=1, max_ppc=64, ws_mb=2048, pq_lut_f16=1,
     ),
+    # BASELINE configs[4] per-shard slice: 1B x 128 SIFT1B-shaped 8-bit SQ
+    # sharded over 8 GPUs = 125M/shard (the HBM-stress config: 16 GB of
+    # packed codes per shard, far beyond the 256 MB LLC). Slab-arena
+    # memory plan keeps peak code memory ~1x + merge_mb.
+    "ivfsq8_125m_d128": dict(
+        type="ivfsq", d=128, n=125_000_000, nlist=16384, m=0, nbits=8,
+        metric=1, nq=10_000, k=10, centers=125_000, sigma=0.5, latent=12,
+        sq_type="8bit", coarse_bf16=1, max_ppc=64, ws_mb=2048,
+    ),
     # scaled-down smoke workload
     "ivfpq_100k_d64": dict(
         type="ivfpq", d=64, n=100_000, nlist=256, m=8, nbits=8,
@@ -241,7 +250,9 @@ def main():
     t0 = time.time()
     eng.add_dev(xb)
     torch.cuda.synchronize()
-    log(f"added {eng.ntotal} in {time.time()-t0:.1f}s")
+    free_b, total_b = torch.cuda.mem_get_info()
+    log(f"added {eng.ntotal} in {time.time()-t0:.1f}s "
+        f"(HBM in use {(total_b-free_b)/2**30:.1f} GiB of {total_b/2**30:.0f})")
 
     # ---- serve through the reference surface (IndexClient) ----
     # The timed step routes through IndexClient.search_dev — the SAME
