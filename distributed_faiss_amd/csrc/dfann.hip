@@ -381,9 +381,16 @@ static void gemm_keys_b(dfann_index *h, const float *A, int64_t Mrows,
     hipLaunchKernelGGL(k_f32_to_bf16, grid1d(Mrows * K), dim3(256), 0, stream,
                        A, Mrows * K, h->ws_bf16a.as<unsigned short>());
     if (K % 8 == 0 && N >= 2048 && Mrows >= 256 && K >= 64) {
-      // big coarse/assign shapes: 256^2-tile glds kernel (512 threads)
+      // big coarse/assign shapes: 256^2-tile glds kernel (512 threads).
+      // DFANN_GEMM_P3=1 selects the BK=32 3-buffer counted-vmcnt ring
+      // (k_gemm_bf16_256_p3) where K % 32 == 0 — experiment knob until
+      // measured on hardware.
       dim3 g2((unsigned)((N + 255) / 256), (unsigned)((Mrows + 255) / 256));
-      hipLaunchKernelGGL(k_gemm_bf16_256, g2, dim3(512), 0, stream,
+      bool p3 = false;
+      if (const char *e = getenv("DFANN_GEMM_P3"))
+        p3 = atoi(e) == 1 && (K % 32) == 0;
+      hipLaunchKernelGGL(p3 ? k_gemm_bf16_256_p3 : k_gemm_bf16_256, g2,
+                         dim3(512), 0, stream,
                          h->ws_bf16a.as<unsigned short>(), B_bf16, keys,
                          (int)Mrows, (int)N, K, K, K, (int)N, qn, bn, mode);
     } else {

@@ -804,6 +804,119 @@ extern "C" __global__ __launch_bounds__(512) void k_gemm_bf16_256(
 }
 
 // ---------------------------------------------------------------------------
+// 3-buffer glds ring variant of the 256² bf16 GEMM (cdna_hip_programming
+// §5 "Pipelining across barriers"): BK=32, three LDS buffers (96 KB), one
+// tile left IN FLIGHT across each barrier with counted s_waitcnt vmcnt —
+// raw s_barrier + lgkmcnt(0) instead of __syncthreads(), whose fence
+// would emit vmcnt(0) and drain the pipeline. This kernel runs at
+// 1 block/CU (the regime where the guide measures the span at +83% over
+// serial staging; the 2-buffer BK=64 k_gemm_bf16_256 is the +40% tier).
+// Requirements: K % 32 == 0 and NON-DIVERGENT staging — out-of-range
+// rows are address-CLAMPED (garbage lands in C rows/cols >= M/N, which
+// the epilogue guard discards) so every wave issues exactly 8 glds per
+// tile and the vmcnt counts are exact.
+// ---------------------------------------------------------------------------
+
+#define GB3_K 32
+
+extern "C" __global__ __launch_bounds__(512) void k_gemm_bf16_256_p3(
+    const unsigned short *__restrict__ A, const unsigned short *__restrict__ B,
+    float *__restrict__ C, int M, int N, int K, int lda, int ldb, int ldc,
+    const float *__restrict__ qn, const float *__restrict__ bn, int mode) {
+  __shared__ unsigned short smem3[3][2][GB2_T][GB3_K];
+  int bi = blockIdx.y * GB2_T;
+  int bj = blockIdx.x * GB2_T;
+  int tid = threadIdx.x;
+  int lane = tid & 63, w = tid >> 6;
+  int wr = (w >> 2) * 128, wc = (w & 3) * 64;
+  f32x4 acc[8][4] = {};
+  int li = lane & 15;
+  int ke = (lane >> 4) * 8;
+
+  // 1024 16-B groups per operand (256 rows x 4), 8 waves x 2 x 64 lanes;
+  // 4 glds per thread per operand-pair iteration => 8 per tile
+#define GLDS3_STAGE(BUF, K0)                                                   \
+  _Pragma("unroll") for (int i = 0; i < 2; ++i) {                              \
+    int g = (w * 2 + i) * 64 + lane;                                           \
+    int r = g >> 2;                                                            \
+    int c8 = (g & 3) * 8;                                                      \
+    int c8s = c8 ^ ((r & 3) << 3); /* BK=32 source-side XOR swizzle */         \
+    int ra = bi + r < M ? bi + r : M - 1;                                      \
+    int rb = bj + r < N ? bj + r : N - 1;                                      \
+    unsigned short *ldst =                                                     \
+        &smem3[BUF][0][0][0] + ((size_t)(w * 2 + i) * 64) * 8;                 \
+    const unsigned short *ga = &A[(size_t)ra * lda + (K0) + c8s];              \
+    const unsigned short *gb = &B[(size_t)rb * ldb + (K0) + c8s];              \
+    __builtin_amdgcn_global_load_lds(                                          \
+        (const __attribute__((address_space(1))) unsigned int *)ga,            \
+        (__attribute__((address_space(3))) unsigned int *)ldst, 16, 0, 0);     \
+    unsigned short *ldstB =                                                    \
+        &smem3[BUF][1][0][0] + ((size_t)(w * 2 + i) * 64) * 8;                 \
+    __builtin_amdgcn_global_load_lds(                                          \
+        (const __attribute__((address_space(1))) unsigned int *)gb,            \
+        (__attribute__((address_space(3))) unsigned int *)ldstB, 16, 0, 0);    \
+  }
+
+#define GLDS3_MFMA(BUF)                                                        \
+  _Pragma("unroll") for (int tj = 0; tj < 4; ++tj) {                           \
+    int rb = wc + tj * 16 + li;                                                \
+    int cbx = ke ^ ((rb & 3) << 3);                                            \
+    bf16x8 b0 = *reinterpret_cast<const bf16x8 *>(&smem3[BUF][1][rb][cbx]);    \
+    _Pragma("unroll") for (int ti = 0; ti < 8; ++ti) {                         \
+      int ra = wr + ti * 16 + li;                                              \
+      int ca = ke ^ ((ra & 3) << 3);                                           \
+      bf16x8 a0 = *reinterpret_cast<const bf16x8 *>(&smem3[BUF][0][ra][ca]);   \
+      acc[ti][tj] =                                                            \
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc[ti][tj],         \
+                                                  0, 0, 0);                    \
+    }                                                                          \
+  }
+
+#define RAW_BARRIER()                                                          \
+  do {                                                                         \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                         \
+    __builtin_amdgcn_s_barrier();                                              \
+  } while (0)
+
+  const int T = K >> 5;  // K % 32 == 0 (host-guaranteed)
+  GLDS3_STAGE(0, 0)
+  if (T > 1) GLDS3_STAGE(1, GB3_K)
+  // buf0 landed (the 4 glds of buf1 may stay in flight)
+  if (T > 1) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  RAW_BARRIER();
+  for (int t = 0; t < T; ++t) {
+    if (t + 2 < T) GLDS3_STAGE((t + 2) % 3, (t + 2) * GB3_K)
+    GLDS3_MFMA(t % 3)
+    if (t + 1 < T) {
+      // next buffer landed; the one after (if staged) stays in flight
+      if (t + 2 < T) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      RAW_BARRIER();
+    }
+  }
+
+  int rrow = (lane >> 4) * 4;
+#pragma unroll
+  for (int ti = 0; ti < 8; ++ti) {
+#pragma unroll
+    for (int tj = 0; tj < 4; ++tj) {
+#pragma unroll
+      for (int rg = 0; rg < 4; ++rg) {
+        int row = bi + wr + ti * 16 + rrow + rg;
+        int col = bj + wc + tj * 16 + li;
+        if (row < M && col < N)
+          C[(size_t)row * ldc + col] =
+              gemm_key(acc[ti][tj][rg], row, col, qn, bn, mode);
+      }
+    }
+  }
+#undef GLDS3_STAGE
+#undef GLDS3_MFMA
+#undef RAW_BARRIER
+}
+
+// ---------------------------------------------------------------------------
 
 extern "C" __global__ void k_assign_init(float *best_v, int *best_i, long long n) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
