@@ -997,6 +997,11 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
         fam_floats = 2 * h->d;
         kern = rk ? (ip ? k_scan_sq8_ip_rk : k_scan_sq8_l2_rk)
                   : (ip ? k_scan_sq8_ip : k_scan_sq8_l2);
+        // DFANN_SCAN_NT=1: non-temporal code-row loads (A/B experiment)
+        if (rk)
+          if (const char *e = getenv("DFANN_SCAN_NT"))
+            if (atoi(e) == 1)
+              kern = ip ? k_scan_sq8_ip_rk_nt : k_scan_sq8_l2_rk_nt;
       } else {
         fam_floats = h->d;
         kern = rk ? (ip ? k_scan_sqf_ip_rk : k_scan_sqf_l2_rk)

@@ -1131,7 +1131,7 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
 }
 
 template <int FAM, bool IS_IP, bool REGSEL, bool PRE = false,
-          bool GLUT = false, bool L16 = false>
+          bool GLUT = false, bool L16 = false, bool NT = false>
 __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
@@ -1353,9 +1353,15 @@ __device__ void ivf_scan_body(
             long long pos = base + (long long)u * NG8 + grp;
             val4[u] = pos < s1;
             int t0 = g8 * 16;
+            // NT: non-temporal (evict-first) loads — the code stream is
+            // read exactly once per step, keep it out of L2's way
             wv4[u] = (val4[u] && t0 < d)
-                         ? *reinterpret_cast<const uint4 *>(
-                               slab_row(codes, rlog, pos, stride) + t0)
+                         ? (NT ? __builtin_nontemporal_load(
+                                     reinterpret_cast<const uint4 *>(
+                                         slab_row(codes, rlog, pos, stride) +
+                                         t0))
+                               : *reinterpret_cast<const uint4 *>(
+                                     slab_row(codes, rlog, pos, stride) + t0))
                          : uint4{0, 0, 0, 0};
           }
 #pragma unroll
@@ -1864,6 +1870,22 @@ extern "C" __global__ __launch_bounds__(256) void k_pq_lut_f16(
     int nq, int nprobe, int d, int m, int dsub, int is_ip, __half *out) {
   DFANN_PQ_LUT_DISPATCH(__half, true, out)
 }
+#define INSTANTIATE_SCAN_NT(NAME, FAM, IS_IP, REGSEL)                          \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,                                                                  \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,         \
+      float *cand_d, unsigned *cand_p, int fam_floats, int fan) {              \
+    ivf_scan_body<FAM, IS_IP, REGSEL, false, false, false, true>(              \
+        q, cent, cb, sq_vmin, sq_scale, probes, keys, codes, off, nq, nprobe,  \
+        d, m, dsub, k, stride, rlog, cand_d, cand_p, fam_floats, nullptr,      \
+        nullptr, nullptr, fan);                                                \
+  }
+INSTANTIATE_SCAN_NT(k_scan_sq8_l2_rk_nt, 2, false, true)
+INSTANTIATE_SCAN_NT(k_scan_sq8_ip_rk_nt, 2, true, true)
+
 INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
 INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)
 INSTANTIATE_SCAN(k_scan_sq8_l2_rk, 2, false, true)

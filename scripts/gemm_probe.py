@@ -41,3 +41,12 @@ for bf16 in (0, 1):
     probe(768, 4096, 65536, bf16)
     probe(768, 65536, 8192, bf16)
     probe(128, 1024, 65536, bf16)
+
+# round-2: BK=32 3-buffer counted-vmcnt ring A/B at the coarse shapes
+print("--- p3 ring A/B (bf16) ---", flush=True)
+for p3 in ("0", "1"):
+    os.environ["DFANN_GEMM_P3"] = p3
+    print(f"DFANN_GEMM_P3={p3}")
+    probe(768, 16384, 10000, 1)   # headline coarse (nlist=16384)
+    probe(768, 65536, 8192, 1)    # old headline coarse (VERDICT item 7 shape)
+    probe(128, 16384, 65536, 1)   # config-5 assign shape
