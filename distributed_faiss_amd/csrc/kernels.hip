@@ -48,6 +48,17 @@ __device__ __forceinline__ uint8_t *slab_row_mut(
 }
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(4))) unsigned uint32x4v;
+
+// 16-B non-temporal (evict-first) load; uint4 is a class type, so go
+// through the ext-vector form the builtin accepts
+__device__ __forceinline__ uint4 nt_load16(const void *p) {
+  uint32x4v v =
+      __builtin_nontemporal_load(reinterpret_cast<const uint32x4v *>(p));
+  union { uint32x4v v; uint4 u; } c;
+  c.v = v;
+  return c.u;
+}
 
 // ---------------------------------------------------------------------------
 // selection machinery: threshold-filtered append + block bitonic compact
@@ -1356,10 +1367,8 @@ __device__ void ivf_scan_body(
             // NT: non-temporal (evict-first) loads — the code stream is
             // read exactly once per step, keep it out of L2's way
             wv4[u] = (val4[u] && t0 < d)
-                         ? (NT ? __builtin_nontemporal_load(
-                                     reinterpret_cast<const uint4 *>(
-                                         slab_row(codes, rlog, pos, stride) +
-                                         t0))
+                         ? (NT ? nt_load16(
+                                     slab_row(codes, rlog, pos, stride) + t0)
                                : *reinterpret_cast<const uint4 *>(
                                      slab_row(codes, rlog, pos, stride) + t0))
                          : uint4{0, 0, 0, 0};
