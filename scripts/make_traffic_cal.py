@@ -21,18 +21,25 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 def main():
     pmc_csv, workload, kregex, nprobe, lut, source = sys.argv[1:7]
-    fetch, write = [], []
+    rows = []
     with open(pmc_csv) as f:
         for row in csv.DictReader(f):
             name = row.get("Kernel_Name", row.get("kernel", ""))
-            if not re.search(kregex, name):
-                continue
-            cname = row.get("Counter_Name", row.get("counter", ""))
-            val = float(row.get("Counter_Value", row.get("value", 0)))
-            if cname == "FETCH_SIZE":
-                fetch.append(val)
-            elif cname == "WRITE_SIZE":
-                write.append(val)
+            if re.search(kregex, name):
+                rows.append(row)
+    # keep only the FULL-BATCH dispatches (largest grid): the bench also
+    # launches small recall-sweep searches that would skew the mean
+    gmax = max(int(r["Grid_Size"]) for r in rows)
+    fetch, write = [], []
+    for row in rows:
+        if int(row["Grid_Size"]) != gmax:
+            continue
+        cname = row.get("Counter_Name", row.get("counter", ""))
+        val = float(row.get("Counter_Value", row.get("value", 0)))
+        if cname == "FETCH_SIZE":
+            fetch.append(val)
+        elif cname == "WRITE_SIZE":
+            write.append(val)
     if not fetch:
         sys.exit(f"no FETCH_SIZE rows matched {kregex!r} in {pmc_csv}")
     # counters are reported in KB per dispatch; gfx950 FETCH_SIZE
