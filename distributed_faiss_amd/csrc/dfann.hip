@@ -14,6 +14,7 @@
 //  * ids are implicit arrival positions (reference quirk).
 
 #include "kernels.hip"
+#include "hnsw.hip"
 
 #include <cstdio>
 #include <cstdlib>
@@ -265,7 +266,7 @@ static std::vector<int64_t> pick_init(int64_t n, int64_t k, uint64_t seed) {
 // index object
 // ---------------------------------------------------------------------------
 
-enum IdxType { T_FLAT = 0, T_IVFFLAT = 1, T_IVFPQ = 2, T_IVFSQ = 3 };
+enum IdxType { T_FLAT = 0, T_IVFFLAT = 1, T_IVFPQ = 2, T_IVFSQ = 3, T_HNSW = 4 };
 enum { M_IP = 0, M_L2 = 1 };
 
 static int round16(int b) { return (b + 15) & ~15; }
@@ -318,6 +319,13 @@ struct dfann_index {
   DevBuf cr_ids, cr_off, id2pos, cr_ids_new;
   std::vector<int64_t> h_off;
   bool dirty = false;
+  // HNSW (type "hnswsq"): graph over non-residual SQ8 codes in csr_arena
+  // (arrival order, never rebuilt). M = h->m, deg0 = 2M.
+  int hnsw_efc = 100;
+  int hnsw_entry = -1, hnsw_maxlevel = -1, hnsw_nslots = 0;
+  std::vector<int32_t> h_levels, h_upslot;
+  DevBuf hn_levels, hn_nbr0, hn_cnt0, hn_upslot, hn_nbrU, hn_cntU;
+  DevBuf hn_req, hn_reqcnt, hn_runoff, hn_u;
   // flat arena
   DevBuf flat;
   // workspace
@@ -596,6 +604,7 @@ static dfann_index *create_from_spec(const std::string &js) {
   else if (t == "ivf_flat") h->type = T_IVFFLAT;
   else if (t == "ivfpq") h->type = T_IVFPQ;
   else if (t == "ivfsq") h->type = T_IVFSQ;
+  else if (t == "hnswsq") h->type = T_HNSW;
   else { delete h; throw std::runtime_error("unknown index type '" + t + "'"); }
   h->d = (int)json_int(js, "dim", 0);
   h->metric = (int)json_int(js, "metric", M_IP);
@@ -618,7 +627,7 @@ static dfann_index *create_from_spec(const std::string &js) {
   if (h->scan_fan < 1) h->scan_fan = 1;
   if (h->scan_fan > 16) h->scan_fan = 16;
   if (h->d <= 0) { delete h; throw std::runtime_error("bad dim"); }
-  if (h->type != T_FLAT && h->nlist <= 0) {
+  if (h->type != T_FLAT && h->type != T_HNSW && h->nlist <= 0) {
     delete h;
     throw std::runtime_error("bad nlist");
   }
@@ -634,6 +643,17 @@ static dfann_index *create_from_spec(const std::string &js) {
     h->code_bytes = h->sq8 ? h->d : 2 * h->d;
   } else if (h->type == T_IVFFLAT) {
     h->code_bytes = 4 * h->d;
+  } else if (h->type == T_HNSW) {
+    // reference index.py:51-60: faiss.IndexHNSWSQ(dim, QT_8bit, store_n)
+    // with L2 asserted; spec m = store_n (link cap M), nprobe = efSearch
+    if (h->metric != M_L2)
+      { delete h; throw std::runtime_error("hnswsq requires L2 (ref index.py:52)"); }
+    if (h->m <= 0) h->m = 128;  // reference store_n default
+    h->hnsw_efc = (int)json_int(js, "ef_construction", 100);
+    if (h->hnsw_efc < 1) h->hnsw_efc = 1;
+    if (h->hnsw_efc > 512) h->hnsw_efc = 512;
+    h->sq8 = true;
+    h->code_bytes = h->d;
   } else {
     h->code_bytes = 4 * h->d;
     h->trained = true;  // flat needs no training
@@ -649,10 +669,36 @@ static dfann_index *create_from_spec(const std::string &js) {
   return h;
 }
 
+static void hnsw_train_ranges(dfann_index *h, int64_t n, const float *x,
+                              hipStream_t stream) {
+  DevBuf mn, mx;
+  mn.ensure((size_t)h->d * 4);
+  mx.ensure((size_t)h->d * 4);
+  h->sq_vmin.ensure((size_t)h->d * 4);
+  h->sq_vdiff.ensure((size_t)h->d * 4);
+  h->sq_scale.ensure((size_t)h->d * 4);
+  hipLaunchKernelGGL(k_minmax_init, grid1d(h->d), dim3(256), 0, stream,
+                     mn.as<unsigned>(), mx.as<unsigned>(), h->d);
+  hipLaunchKernelGGL(k_minmax_dims, grid1d(n * h->d), dim3(256), 0, stream, x,
+                     n, h->d, mn.as<unsigned>(), mx.as<unsigned>());
+  hipLaunchKernelGGL(k_minmax_decode, grid1d(h->d), dim3(256), 0, stream,
+                     mn.as<unsigned>(), mx.as<unsigned>(), h->d,
+                     h->sq_vmin.as<float>(), h->sq_vdiff.as<float>(),
+                     h->sq_scale.as<float>());
+  HIP_CHECK(hipStreamSynchronize(stream));
+}
+
 static void train_impl(dfann_index *h, int64_t n, const float *x,
                        hipStream_t stream) {
   if (h->type == T_FLAT) { h->trained = true; return; }
   if (h->trained) return;  // faiss semantics: re-train of trained is a no-op here
+  if (h->type == T_HNSW) {
+    // non-residual SQ8 codec: per-dim ranges over the raw training set
+    // (faiss IndexHNSWSQ trains its storage IndexScalarQuantizer)
+    hnsw_train_ranges(h, n, x, stream);
+    h->trained = true;
+    return;
+  }
   h->centroids.ensure((size_t)h->nlist * h->d * 4);
   h->cnorm.ensure((size_t)h->nlist * 4);
   trace_point("train:begin", stream);
@@ -873,10 +919,14 @@ static void encode_chunk(dfann_index *h, int64_t n, const float *x,
   HIP_CHECK(hipGetLastError());
 }
 
+static void hnsw_add(dfann_index *h, int64_t n, const float *x,
+                     hipStream_t stream);
+
 static void add_impl(dfann_index *h, int64_t n, const float *x,
                      hipStream_t stream) {
   if (!h->trained) throw std::runtime_error("add on untrained index");
   if (n == 0) return;
+  if (h->type == T_HNSW) { hnsw_add(h, n, x, stream); return; }
   if (h->type == T_FLAT) {
     h->flat.grow_keep(((size_t)h->ntotal + n) * h->d * 4,
                       (size_t)h->ntotal * h->d * 4);
@@ -916,6 +966,189 @@ static void add_impl(dfann_index *h, int64_t n, const float *x,
 
 static void finalize_csr(dfann_index *h, hipStream_t stream) {
   rebuild_csr(h, stream);
+}
+
+
+// ---------------------------------------------------------------------------
+// HNSW host orchestration (type "hnswsq")
+// ---------------------------------------------------------------------------
+
+// deterministic level draw (splitmix64; clamped). The oracle does not
+// replicate this (graph parity is via the engine's own dump).
+static int hnsw_draw_level(uint64_t seed, int64_t id, int M) {
+  SplitMix64 rng(seed ^ (uint64_t)(id * 0x9E3779B97F4A7C15ULL + 0x51ED2701ULL));
+  double u = ((double)(rng.next() >> 11) + 1.0) * (1.0 / 9007199254740992.0);
+  double mL = 1.0 / log((double)M);
+  int l = (int)floor(-log(u) * mL);
+  if (l < 0) l = 0;
+  if (l > HNSW_MAXL) l = HNSW_MAXL;
+  return l;
+}
+
+static void hnsw_add(dfann_index *h, int64_t n, const float *x,
+                     hipStream_t stream) {
+  const int M = h->m, deg0 = 2 * h->m;
+  const int64_t n0 = h->ntotal, ntot = n0 + n;
+  // 1) encode rows [n0, ntot) into the code arena (arrival order; the
+  //    HNSW arena is never rebuilt — ids ARE row positions)
+  h->csr_arena.ensure_rows(ntot);
+  {
+    const int64_t CH = std::max<int64_t>(
+        65536, ((int64_t)h->ws_mb << 20) / std::max(4 * h->d, 1));
+    for (int64_t s0 = 0; s0 < n; s0 += CH) {
+      int64_t c = std::min(CH, n - s0);
+      hipLaunchKernelGGL(k_sq_encode, grid1d(c * h->d), dim3(256), 0, stream,
+                         x + s0 * h->d, h->sq_vmin.as<float>(),
+                         h->sq_vdiff.as<float>(), c, h->d, h->stride, 0,
+                         h->csr_arena.rlog, n0 + s0,
+                         h->csr_arena.dev_table_mut(stream));
+    }
+    HIP_CHECK(hipGetLastError());
+  }
+  // 2) levels + upper slots (host, deterministic)
+  h->h_levels.resize(ntot);
+  h->h_upslot.resize(ntot);
+  int old_nslots = h->hnsw_nslots;
+  for (int64_t i = n0; i < ntot; ++i) {
+    int lv = hnsw_draw_level(h->seed, i, M);
+    h->h_levels[i] = lv;
+    h->h_upslot[i] = lv > 0 ? h->hnsw_nslots++ : -1;
+  }
+  // 3) grow device graph arrays (grow_keep preserves; new ranges zeroed
+  //    or uploaded). Graph memory is deg0*4 B/node — the dominant HNSW
+  //    cost, same as faiss's links storage.
+  h->hn_levels.grow_keep((size_t)ntot * 4, (size_t)n0 * 4);
+  HIP_CHECK(hipMemcpyAsync(h->hn_levels.as<int>() + n0,
+                           h->h_levels.data() + n0, (size_t)n * 4,
+                           hipMemcpyHostToDevice, stream));
+  h->hn_upslot.grow_keep((size_t)ntot * 4, (size_t)n0 * 4);
+  HIP_CHECK(hipMemcpyAsync(h->hn_upslot.as<int>() + n0,
+                           h->h_upslot.data() + n0, (size_t)n * 4,
+                           hipMemcpyHostToDevice, stream));
+  h->hn_nbr0.grow_keep((size_t)ntot * deg0 * 4, (size_t)n0 * deg0 * 4);
+  h->hn_cnt0.grow_keep((size_t)ntot * 4, (size_t)n0 * 4);
+  HIP_CHECK(hipMemsetAsync(h->hn_cnt0.as<int>() + n0, 0, (size_t)n * 4,
+                           stream));
+  if (h->hnsw_nslots > old_nslots) {
+    h->hn_nbrU.grow_keep((size_t)h->hnsw_nslots * HNSW_MAXL * M * 4,
+                         (size_t)old_nslots * HNSW_MAXL * M * 4);
+    h->hn_cntU.grow_keep((size_t)h->hnsw_nslots * HNSW_MAXL * 4,
+                         (size_t)old_nslots * HNSW_MAXL * 4);
+    HIP_CHECK(hipMemsetAsync(
+        h->hn_cntU.as<int>() + (size_t)old_nslots * HNSW_MAXL, 0,
+        (size_t)(h->hnsw_nslots - old_nslots) * HNSW_MAXL * 4, stream));
+  }
+  // 4) wave insertion over frozen snapshots
+  const int64_t WMAX = 4096;
+  const int64_t req_cap = WMAX * M * (HNSW_MAXL + 1);
+  h->hn_req.ensure((size_t)req_cap * 16);
+  h->hn_reqcnt.ensure(4);
+  h->hn_u.ensure((size_t)WMAX * h->d * 4);
+  size_t lds_ins = HNSW_LDS_INS(h->d, 256);
+  size_t lds_apply = HNSW_LDS_APPLY(h->d);
+  std::vector<int> hreq;
+  std::vector<int64_t> order;
+  std::vector<int> runoff;
+  int64_t w0 = n0;
+  while (w0 < ntot) {
+    if (h->hnsw_entry < 0) {  // very first point
+      h->hnsw_entry = 0;
+      h->hnsw_maxlevel = h->h_levels[0];
+    }
+    int64_t W = w0 == 0 ? 1 : std::min<int64_t>(std::min(WMAX, w0), ntot - w0);
+    hipLaunchKernelGGL(k_hnsw_prep, grid1d(W * h->d), dim3(256), 0, stream,
+                       x + (w0 - n0) * h->d, h->sq_vmin.as<float>(),
+                       h->sq_scale.as<float>(), W, h->d, h->hn_u.as<float>());
+    HIP_CHECK(hipMemsetAsync(h->hn_reqcnt.p, 0, 4, stream));
+    hipLaunchKernelGGL(k_hnsw_insert, dim3((unsigned)W), dim3(256), lds_ins,
+                       stream, h->hn_u.as<float>(), h->sq_scale.as<float>(),
+                       h->csr_arena.dev_table(stream), h->csr_arena.rlog,
+                       h->stride, h->d, h->hn_levels.as<int>(),
+                       h->hn_nbr0.as<int>(), h->hn_cnt0.as<int>(),
+                       h->hn_upslot.as<int>(), h->hn_nbrU.as<int>(),
+                       h->hn_cntU.as<int>(), deg0, M, w0, h->hnsw_entry,
+                       h->hnsw_maxlevel, w0, (int)W, h->hnsw_efc,
+                       h->hn_req.as<int>(), h->hn_reqcnt.as<int>(),
+                       (int)req_cap);
+    HIP_CHECK(hipGetLastError());
+    int rcnt = 0;
+    HIP_CHECK(hipMemcpyAsync(&rcnt, h->hn_reqcnt.p, 4, hipMemcpyDeviceToHost,
+                             stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    if (rcnt > req_cap)
+      throw std::runtime_error("hnsw: reverse-link buffer overflow");
+    if (rcnt > 0) {
+      // sort requests by (dst, level, src) -> deterministic application
+      hreq.resize((size_t)rcnt * 4);
+      HIP_CHECK(hipMemcpy(hreq.data(), h->hn_req.p, (size_t)rcnt * 16,
+                          hipMemcpyDeviceToHost));
+      order.resize(rcnt);
+      for (int i = 0; i < rcnt; ++i) order[i] = i;
+      std::sort(order.begin(), order.end(), [&](int64_t a, int64_t b) {
+        const int *ra = &hreq[(size_t)a * 4], *rb = &hreq[(size_t)b * 4];
+        if (ra[0] != rb[0]) return ra[0] < rb[0];
+        if (ra[2] != rb[2]) return ra[2] < rb[2];
+        return ra[1] < rb[1];
+      });
+      std::vector<int> sorted((size_t)rcnt * 4);
+      runoff.clear();
+      for (int i = 0; i < rcnt; ++i) {
+        const int *r = &hreq[(size_t)order[i] * 4];
+        if (i == 0 || r[0] != sorted[(size_t)(i - 1) * 4] ||
+            r[2] != sorted[(size_t)(i - 1) * 4 + 2])
+          runoff.push_back(i);
+        memcpy(&sorted[(size_t)i * 4], r, 16);
+      }
+      runoff.push_back(rcnt);
+      int n_runs = (int)runoff.size() - 1;
+      HIP_CHECK(hipMemcpy(h->hn_req.p, sorted.data(), (size_t)rcnt * 16,
+                          hipMemcpyHostToDevice));
+      h->hn_runoff.ensure(runoff.size() * 4);
+      HIP_CHECK(hipMemcpy(h->hn_runoff.p, runoff.data(), runoff.size() * 4,
+                          hipMemcpyHostToDevice));
+      hipLaunchKernelGGL(k_hnsw_apply, dim3((unsigned)n_runs), dim3(256),
+                         lds_apply, stream, h->sq_scale.as<float>(),
+                         h->csr_arena.dev_table(stream), h->csr_arena.rlog,
+                         h->stride, h->d, h->hn_nbr0.as<int>(),
+                         h->hn_cnt0.as<int>(), h->hn_upslot.as<int>(),
+                         h->hn_nbrU.as<int>(), h->hn_cntU.as<int>(), deg0, M,
+                         h->hn_req.as<int>(), h->hn_runoff.as<int>(), n_runs);
+      HIP_CHECK(hipGetLastError());
+      HIP_CHECK(hipStreamSynchronize(stream));
+    }
+    // entry update: highest new level (lowest id on ties) beats the old
+    for (int64_t i = w0; i < w0 + W; ++i)
+      if (h->h_levels[i] > h->hnsw_maxlevel) {
+        h->hnsw_maxlevel = h->h_levels[i];
+        h->hnsw_entry = (int)i;
+      }
+    w0 += W;
+  }
+  h->csr_arena.rows = ntot;
+  h->ntotal = ntot;
+}
+
+static void hnsw_search(dfann_index *h, int64_t nq, const float *q, int k,
+                        float *D, int64_t *I, hipStream_t stream) {
+  int ef = std::max(h->nprobe, k);
+  if (ef > 512) ef = 512;
+  h->hn_u.ensure((size_t)nq * h->d * 4);
+  hipLaunchKernelGGL(k_hnsw_prep, grid1d(nq * h->d), dim3(256), 0, stream, q,
+                     h->sq_vmin.as<float>(), h->sq_scale.as<float>(), nq, h->d,
+                     h->hn_u.as<float>());
+  TimingEv e;
+  if (h->timing) e = h->ev_begin(stream);
+  hipLaunchKernelGGL(k_hnsw_search, dim3((unsigned)nq), dim3(256),
+                     HNSW_LDS(h->d, 256), stream, h->hn_u.as<float>(),
+                     h->sq_scale.as<float>(), h->csr_arena.dev_table(stream),
+                     h->csr_arena.rlog, h->stride, h->d,
+                     h->hn_levels.as<int>(), h->hn_nbr0.as<int>(),
+                     h->hn_cnt0.as<int>(), h->hn_upslot.as<int>(),
+                     h->hn_nbrU.as<int>(), h->hn_cntU.as<int>(), 2 * h->m,
+                     h->m, h->ntotal, h->hnsw_entry, h->hnsw_maxlevel, nq, ef,
+                     k, D, I);
+  if (h->timing) h->ev_end(e, stream, h->ev_scan);
+  HIP_CHECK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------------
@@ -1284,6 +1517,7 @@ static void search_impl(dfann_index *h, int64_t nq, const float *q, int k,
   if (nq == 0) return;
   if (h->type == T_FLAT) { flat_search_impl(h, nq, q, k, D, I, stream); return; }
   if (h->ntotal == 0) { pad_fill(h, nq, k, D, I, stream); return; }
+  if (h->type == T_HNSW) { hnsw_search(h, nq, q, k, D, I, stream); return; }
   finalize_csr(h, stream);
   int nprobe = std::min(h->nprobe, h->nlist);
   if (nprobe > 512) {
@@ -1392,6 +1626,7 @@ extern "C" int dfann_search_reconstruct(dfann_index *h, int64_t nq,
   if (h->type == T_FLAT) { rtype = 0; flat_src = h->flat.as<float>(); }
   else if (h->type == T_IVFFLAT) rtype = 0;
   else if (h->type == T_IVFPQ) rtype = 2;
+  else if (h->type == T_HNSW) rtype = 5;
   else rtype = h->sq8 ? 3 : 4;
   hipLaunchKernelGGL(k_reconstruct, dim3((unsigned)(nq * k)), dim3(64), 0,
                      (hipStream_t)stream, I_dev, nq, k, rtype, h->d, h->m,
@@ -1423,6 +1658,9 @@ extern "C" int dfann_get_centroids(dfann_index *h, float *out_host) {
   API_BEGIN
   if (h->type == T_FLAT)
     throw std::runtime_error("'flat' index has no quantizer");
+  if (h->type == T_HNSW)
+    throw std::runtime_error(
+        "hnswsq index has no quantizer (reference AttributeError)");
   if (!h->trained) throw std::runtime_error("index not trained");
   HIP_CHECK(hipMemcpy(out_host, h->centroids.p, (size_t)h->nlist * h->d * 4,
                       hipMemcpyDeviceToHost));
@@ -1441,6 +1679,8 @@ extern "C" int dfann_get_lists(dfann_index *h, int64_t *off_host,
                                int64_t *ids_host, uint8_t *codes_host) {
   API_BEGIN
   if (h->type == T_FLAT) throw std::runtime_error("flat index has no lists");
+  if (h->type == T_HNSW)
+    throw std::runtime_error("hnswsq index has no inverted lists");
   finalize_csr(h, 0);
   memcpy(off_host, h->h_off.data(), (size_t)(h->nlist + 1) * 8);
   if (h->ntotal) {
@@ -1456,7 +1696,7 @@ extern "C" int dfann_code_stride(dfann_index *h) { return h->stride; }
 extern "C" int dfann_get_sq_params(dfann_index *h, float *vmin_host,
                                    float *vdiff_host) {
   API_BEGIN
-  if (!(h->type == T_IVFSQ && h->sq8))
+  if (!((h->type == T_IVFSQ || h->type == T_HNSW) && h->sq8))
     throw std::runtime_error("not an 8-bit ivfsq index");
   HIP_CHECK(hipMemcpy(vmin_host, h->sq_vmin.p, (size_t)h->d * 4,
                       hipMemcpyDeviceToHost));
@@ -1571,15 +1811,35 @@ extern "C" int dfann_save(dfann_index *h, const char *path) {
     fwrite_chk(&tr, 1, f);
     fwrite_chk(&h->ntotal, 8, f);
     if (h->trained && h->type != T_FLAT) {
-      dump_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
+      if (h->type != T_HNSW)
+        dump_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
       if (h->type == T_IVFPQ)
         dump_dev(f, h->codebooks, (size_t)h->m * 256 * h->dsub * 4);
-      if (h->type == T_IVFSQ && h->sq8) {
+      if ((h->type == T_IVFSQ || h->type == T_HNSW) && h->sq8) {
         dump_dev(f, h->sq_vmin, (size_t)h->d * 4);
         dump_dev(f, h->sq_vdiff, (size_t)h->d * 4);
       }
     }
-    if (h->type == T_FLAT) {
+    if (h->type == T_HNSW) {
+      // codes (arrival order) + the graph
+      const int64_t CHROWS =
+          std::max<int64_t>(1, ((int64_t)64 << 20) / h->stride);
+      std::vector<char> tmp((size_t)CHROWS * h->stride);
+      for (int64_t r = 0; r < h->ntotal; r += CHROWS) {
+        int64_t c = std::min(CHROWS, h->ntotal - r);
+        h->csr_arena.copy_rows_to_host(tmp.data(), r, c);
+        fwrite_chk(tmp.data(), (size_t)c * h->stride, f);
+      }
+      int32_t meta[4] = {h->hnsw_entry, h->hnsw_maxlevel, h->hnsw_nslots,
+                         h->hnsw_efc};
+      fwrite_chk(meta, sizeof(meta), f);
+      fwrite_chk(h->h_levels.data(), (size_t)h->ntotal * 4, f);
+      fwrite_chk(h->h_upslot.data(), (size_t)h->ntotal * 4, f);
+      dump_dev(f, h->hn_cnt0, (size_t)h->ntotal * 4);
+      dump_dev(f, h->hn_nbr0, (size_t)h->ntotal * 2 * h->m * 4);
+      dump_dev(f, h->hn_cntU, (size_t)h->hnsw_nslots * HNSW_MAXL * 4);
+      dump_dev(f, h->hn_nbrU, (size_t)h->hnsw_nslots * HNSW_MAXL * h->m * 4);
+    } else if (h->type == T_FLAT) {
       dump_dev(f, h->flat, (size_t)h->ntotal * h->d * 4);
     } else if (h->ntotal) {
       fwrite_chk(h->h_off.data(), (size_t)(h->nlist + 1) * 8, f);
@@ -1624,7 +1884,17 @@ extern "C" int dfann_load(const char *path, dfann_index **out) {
     fread_chk(&tr, 1, f);
     fread_chk(&h->ntotal, 8, f);
     h->trained = tr != 0;
-    if (h->trained && h->type != T_FLAT) {
+    if (h->trained && h->type == T_HNSW) {
+      load_dev(f, h->sq_vmin, (size_t)h->d * 4);
+      load_dev(f, h->sq_vdiff, (size_t)h->d * 4);
+      std::vector<float> vd(h->d), sc(h->d);
+      HIP_CHECK(hipMemcpy(vd.data(), h->sq_vdiff.p, (size_t)h->d * 4,
+                          hipMemcpyDeviceToHost));
+      for (int t = 0; t < h->d; ++t) sc[t] = vd[t] / 255.0f;
+      h->sq_scale.ensure((size_t)h->d * 4);
+      HIP_CHECK(hipMemcpy(h->sq_scale.p, sc.data(), (size_t)h->d * 4,
+                          hipMemcpyHostToDevice));
+    } else if (h->trained && h->type != T_FLAT) {
       load_dev(f, h->centroids, (size_t)h->nlist * h->d * 4);
       h->cnorm.ensure((size_t)h->nlist * 4);
       rownorms(h->centroids.as<float>(), h->nlist, h->d, h->cnorm.as<float>(), 0);
@@ -1643,7 +1913,41 @@ extern "C" int dfann_load(const char *path, dfann_index **out) {
                             hipMemcpyHostToDevice));
       }
     }
-    if (h->type == T_FLAT) {
+    if (h->type == T_HNSW) {
+      if (h->ntotal) {
+        const int64_t CHROWS =
+            std::max<int64_t>(1, ((int64_t)64 << 20) / h->stride);
+        std::vector<char> tmp((size_t)CHROWS * h->stride);
+        for (int64_t r = 0; r < h->ntotal; r += CHROWS) {
+          int64_t c = std::min(CHROWS, h->ntotal - r);
+          fread_chk(tmp.data(), (size_t)c * h->stride, f);
+          h->csr_arena.copy_rows_from_host(tmp.data(), r, c);
+        }
+        h->csr_arena.rows = h->ntotal;
+        int32_t meta[4];
+        fread_chk(meta, sizeof(meta), f);
+        h->hnsw_entry = meta[0];
+        h->hnsw_maxlevel = meta[1];
+        h->hnsw_nslots = meta[2];
+        h->hnsw_efc = meta[3];
+        h->h_levels.resize(h->ntotal);
+        h->h_upslot.resize(h->ntotal);
+        fread_chk(h->h_levels.data(), (size_t)h->ntotal * 4, f);
+        fread_chk(h->h_upslot.data(), (size_t)h->ntotal * 4, f);
+        h->hn_levels.ensure((size_t)h->ntotal * 4);
+        HIP_CHECK(hipMemcpy(h->hn_levels.p, h->h_levels.data(),
+                            (size_t)h->ntotal * 4, hipMemcpyHostToDevice));
+        h->hn_upslot.ensure((size_t)h->ntotal * 4);
+        HIP_CHECK(hipMemcpy(h->hn_upslot.p, h->h_upslot.data(),
+                            (size_t)h->ntotal * 4, hipMemcpyHostToDevice));
+        load_dev(f, h->hn_cnt0, (size_t)h->ntotal * 4);
+        load_dev(f, h->hn_nbr0, (size_t)h->ntotal * 2 * h->m * 4);
+        load_dev(f, h->hn_cntU, (size_t)h->hnsw_nslots * HNSW_MAXL * 4);
+        load_dev(f, h->hn_nbrU,
+                 (size_t)h->hnsw_nslots * HNSW_MAXL * h->m * 4);
+        HIP_CHECK(hipDeviceSynchronize());
+      }
+    } else if (h->type == T_FLAT) {
       load_dev(f, h->flat, (size_t)h->ntotal * h->d * 4);
     } else if (h->ntotal) {
       h->h_off.resize(h->nlist + 1);
@@ -1684,6 +1988,47 @@ extern "C" int dfann_load(const char *path, dfann_index **out) {
   }
   fclose(f);
   *out = h;
+  API_END
+}
+
+// ---------------------------------------------------------------------------
+// HNSW introspection (test plumbing: graph determinism + oracle-shared
+// search parity — DESIGN.md §hnsw)
+// ---------------------------------------------------------------------------
+
+extern "C" int dfann_hnsw_info(dfann_index *h, int64_t out[6]) {
+  API_BEGIN
+  if (h->type != T_HNSW) throw std::runtime_error("not an hnswsq index");
+  out[0] = h->m;
+  out[1] = 2 * h->m;
+  out[2] = h->hnsw_nslots;
+  out[3] = h->hnsw_entry;
+  out[4] = h->hnsw_maxlevel;
+  out[5] = h->hnsw_efc;
+  API_END
+}
+
+extern "C" int dfann_hnsw_dump(dfann_index *h, int32_t *levels_host,
+                               int32_t *cnt0_host, int32_t *nbr0_host,
+                               int32_t *upslot_host, int32_t *cntU_host,
+                               int32_t *nbrU_host) {
+  API_BEGIN
+  if (h->type != T_HNSW) throw std::runtime_error("not an hnswsq index");
+  int64_t n = h->ntotal;
+  memcpy(levels_host, h->h_levels.data(), (size_t)n * 4);
+  memcpy(upslot_host, h->h_upslot.data(), (size_t)n * 4);
+  HIP_CHECK(hipMemcpy(cnt0_host, h->hn_cnt0.p, (size_t)n * 4,
+                      hipMemcpyDeviceToHost));
+  HIP_CHECK(hipMemcpy(nbr0_host, h->hn_nbr0.p, (size_t)n * 2 * h->m * 4,
+                      hipMemcpyDeviceToHost));
+  if (h->hnsw_nslots) {
+    HIP_CHECK(hipMemcpy(cntU_host, h->hn_cntU.p,
+                        (size_t)h->hnsw_nslots * HNSW_MAXL * 4,
+                        hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemcpy(nbrU_host, h->hn_nbrU.p,
+                        (size_t)h->hnsw_nslots * HNSW_MAXL * h->m * 4,
+                        hipMemcpyDeviceToHost));
+  }
   API_END
 }
 

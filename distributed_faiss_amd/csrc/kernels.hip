@@ -2364,6 +2364,12 @@ extern "C" __global__ __launch_bounds__(64) void k_reconstruct(
     for (int t = threadIdx.x; t < d; t += blockDim.x) out[t] = src[t];
     return;
   }
+  if (type == 5) {  // hnsw: non-residual SQ8, codes indexed by id
+    const uint8_t *cp5 = slab_row(codes, rlog, id, stride);
+    for (int t = threadIdx.x; t < d; t += blockDim.x)
+      out[t] = vmin[t] + ((float)cp5[t] + 0.5f) * scale[t];
+    return;
+  }
   unsigned pos = id2pos[id];
   // binary search list containing pos
   int lo = 0, hi = nlist;

@@ -74,10 +74,24 @@ def _builder_spec(cfg: IndexCfg) -> dict:
         # reference index.py:63-68 — QT_fp16
         spec.update(type="ivfsq", nlist=int(cfg.centroids), sq_type="fp16")
         return spec
-    if t in ("hnswsq", "ivf_gpu"):
+    if t == "hnswsq":
+        # reference index.py:51-60: faiss.IndexHNSWSQ(dim, QT_8bit,
+        # store_n) — L2 asserted; efSearch = cfg.nprobe at build,
+        # efConstruction from cfg.extra. Deviation (documented,
+        # DESIGN.md §hnsw): our engine honors later set_nprobe as
+        # efSearch, where the reference's set_nprobe is a silent no-op
+        # on an HNSW index (faiss_index.nprobe is not efSearch).
+        assert cfg.get_metric() == 1, "hnsw is supposed to work with L2 sim space"
+        spec.update(
+            type="hnswsq",
+            m=int(cfg.extra.get("store_n", 128)),
+            ef_construction=int(cfg.extra.get("ef_construction", 100)),
+        )
+        return spec
+    if t == "ivf_gpu":
         raise NotImplementedError(
-            f"index_builder_type={t!r} is out of scope for the MI355X hot-path build "
-            "(SURVEY.md §2: graph index / faiss-GPU wrapper)"
+            "index_builder_type='ivf_gpu' is out of scope (the faiss-GPU "
+            "wrapper is exactly what this engine replaces, SURVEY.md §2)"
         )
     raise KeyError(f"unknown index_builder_type {t!r}")
 

@@ -94,6 +94,8 @@ def load_lib():
                                      c.c_void_p, c.c_void_p]
     lib.dfann_set_timing.argtypes = [c.c_void_p, c.c_int]
     lib.dfann_get_timing.argtypes = [c.c_void_p, P(DfannTiming)]
+    lib.dfann_hnsw_info.argtypes = [c.c_void_p, c.c_void_p]
+    lib.dfann_hnsw_dump.argtypes = [c.c_void_p] + [c.c_void_p] * 6
     lib.dfann_last_error.restype = c.c_char_p
     _LIB = lib
     return lib
@@ -315,6 +317,37 @@ class HipEngine:
             cb.ctypes.data_as(ctypes.c_void_p) if cb is not None else None,
             vm.ctypes.data_as(ctypes.c_void_p) if vm is not None else None,
             vd.ctypes.data_as(ctypes.c_void_p) if vd is not None else None))
+
+    # -- hnsw introspection (test plumbing) -------------------------------
+
+    HNSW_MAXL = 8
+
+    def hnsw_info(self):
+        """(M, deg0, nslots, entry, maxlevel, ef_construction)."""
+        out = np.empty(6, dtype=np.int64)
+        _check(self.lib, self.lib.dfann_hnsw_info(
+            self.h, out.ctypes.data_as(ctypes.c_void_p)))
+        return tuple(int(v) for v in out)
+
+    def hnsw_dump(self):
+        """Full graph to host: dict of levels/cnt0/nbr0/upslot/cntU/nbrU
+        + info tuple. Deterministic given (data, seed, wave schedule)."""
+        M, deg0, nslots, entry, maxlevel, efc = self.hnsw_info()
+        n = self.ntotal
+        L = self.HNSW_MAXL
+        levels = np.empty(n, dtype=np.int32)
+        cnt0 = np.empty(n, dtype=np.int32)
+        nbr0 = np.empty((n, deg0), dtype=np.int32)
+        upslot = np.empty(n, dtype=np.int32)
+        cntU = np.empty((max(nslots, 1), L), dtype=np.int32)
+        nbrU = np.empty((max(nslots, 1), L, M), dtype=np.int32)
+        _check(self.lib, self.lib.dfann_hnsw_dump(
+            self.h, *(a.ctypes.data_as(ctypes.c_void_p)
+                      for a in (levels, cnt0, nbr0, upslot, cntU, nbrU))))
+        return {"levels": levels, "cnt0": cnt0, "nbr0": nbr0,
+                "upslot": upslot, "cntU": cntU, "nbrU": nbrU,
+                "M": M, "deg0": deg0, "nslots": nslots, "entry": entry,
+                "maxlevel": maxlevel, "efc": efc}
 
     # -- persistence / timing ---------------------------------------------
 

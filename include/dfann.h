@@ -144,6 +144,25 @@ int dfann_merge_topk(int64_t nq, int S, int k, const float *D_dev,
                      const int64_t *I_dev, int maximize, float *Dout_dev,
                      int64_t *Iout_dev, dfann_stream stream);
 
+/* --- hnswsq (replaces faiss.IndexHNSWSQ at ref index.py:51-60) --------
+ * spec: {"type": "hnswsq", "dim", "metric": 1 (L2 only — the reference
+ * asserts), "m": store_n (link cap M; level-0 cap 2M), "ef_construction",
+ * "nprobe" (= efSearch; honored dynamically — deviation: the
+ * reference's set_nprobe is a silent no-op on HNSW), "seed"}.
+ * Build is a BATCHED wave insertion over frozen snapshots with reverse
+ * links applied in sorted order — deterministic given (data, seed, wave
+ * schedule), but a DIFFERENT graph than faiss's sequential insertion
+ * (quality gated by recall tests). Search parity is
+ * pinned by the oracle restatement over the dumped graph. ------------- */
+
+/* info: out = {M, deg0, nslots, entry, maxlevel, ef_construction} */
+int dfann_hnsw_info(dfann_index *h, int64_t out[6]);
+/* dump the graph to host: levels (n i32), cnt0 (n), nbr0 (n x 2M),
+ * upslot (n), cntU (nslots x 8), nbrU (nslots x 8 x M) */
+int dfann_hnsw_dump(dfann_index *h, int32_t *levels_host, int32_t *cnt0_host,
+                    int32_t *nbr0_host, int32_t *upslot_host,
+                    int32_t *cntU_host, int32_t *nbrU_host);
+
 /* --- kernel timing for the roofline harness (bench.py) ---------------- */
 
 typedef struct dfann_timing {

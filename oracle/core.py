@@ -805,3 +805,167 @@ def aggregate_results(shard_D, shard_I_meta, topk, maximize):
         Dout[i] = allD[i, order]
         meta_out.append([flat_meta[synth[i, o]] for o in order])
     return Dout, meta_out
+
+
+# ---------------------------------------------------------------------------
+# HNSW search restatement (shared-graph parity tier — DESIGN.md §hnsw).
+#
+# Mirrors csrc/hnsw.hip k_hnsw_search OP-FOR-OP over a graph imported
+# from the engine (hip_engine.HipEngine.hnsw_dump): the 8-lane partial
+# sums + fixed butterfly of the distance, the 8192-slot visited hash
+# with probe cap 32, the Sel frontier semantics (top-ef of everything
+# appended, expansion = first unexpanded in (dist, flagged-id) order),
+# and the two-stage greedy argmin tie structure. Distances are bitwise
+# equal and result (D, I) exactly equal for any graph. The BUILD side is
+# not restated (the batched wave insertion is engine-defined; build
+# quality is gated by recall property tests and build determinism by the
+# dump — oracle/__init__.py pinning notes).
+# ---------------------------------------------------------------------------
+
+HNSW_HASH = 8192
+HNSW_PROBES = 32
+HNSW_MAXL = 8
+_FLT_MAX = np.float32(3.402823466e+38)
+
+
+class OracleHNSWSearch:
+    def __init__(self, d, graph, vmin, vdiff, codes):
+        self.d = d
+        self.g = graph  # dict from hnsw_dump
+        self.vmin = vmin.astype(np.float32)
+        self.vdiff = vdiff.astype(np.float32)
+        self.scale = (self.vdiff / np.float32(255.0)).astype(np.float32)
+        self.codes = codes  # (n, d) uint8, non-residual SQ8
+
+    @staticmethod
+    def encode(x, vmin, vdiff):
+        """k_sq_encode restated (trunc toward zero, f32 ops)."""
+        x = x.astype(np.float32)
+        xi = (x - vmin[None, :].astype(np.float32)) / vdiff[None, :].astype(np.float32)
+        c = (np.float32(255.0) * xi).astype(np.int32)  # trunc toward zero
+        return np.clip(c, 0, 255).astype(np.uint8)
+
+    def _dist_q(self, u, cid):
+        # 8-lane partials (lane g8 covers bytes [g8*16,g8*16+16) stride
+        # 128, sequential within), then the fixed xor-butterfly 4,2,1
+        d = self.d
+        s = self.scale
+        row = self.codes[cid]
+        parts = np.zeros(8, dtype=np.float32)
+        for g8 in range(8):
+            acc = np.float32(0.0)
+            t0 = g8 * 16
+            while t0 < d:
+                for b in range(16):
+                    t = t0 + b
+                    if t < d:
+                        diff = np.float32(u[t] - np.float32(row[t]) * s[t])
+                        acc = np.float32(acc + np.float32(diff * diff))
+                t0 += 128
+            parts[g8] = acc
+        b1 = np.array([np.float32(parts[i] + parts[i ^ 4]) for i in range(8)],
+                      dtype=np.float32)
+        b2 = np.array([np.float32(b1[i] + b1[i ^ 2]) for i in range(8)],
+                      dtype=np.float32)
+        b3 = np.float32(b2[0] + b2[1])
+        return b3
+
+    def _nbrs(self, node, level):
+        g = self.g
+        if level == 0:
+            c = int(g["cnt0"][node])
+            return g["nbr0"][node, :c]
+        slot = int(g["upslot"][node])
+        c = int(g["cntU"][slot, level - 1])
+        return g["nbrU"][slot, level - 1, :c]
+
+    def _greedy(self, u, cur, cur_d, level):
+        NG = 32  # blockDim 256 / 8
+        while True:
+            nb = self._nbrs(cur, level)
+            cnt = len(nb)
+            # two-stage argmin: per-group best over passes, then groups
+            gd = np.full(NG, _FLT_MAX, dtype=np.float32)
+            gi = np.full(NG, -1, dtype=np.int64)
+            for c0 in range(0, cnt, NG):
+                for grp in range(NG):
+                    ci = c0 + grp
+                    if ci < cnt:
+                        nid = int(nb[ci])
+                        dd = self._dist_q(u, nid)
+                        if dd < gd[grp] or (dd == gd[grp] and nid < gi[grp]):
+                            gd[grp] = dd
+                            gi[grp] = nid
+            bd, bi = gd[0], gi[0]
+            for i in range(1, NG):
+                if gd[i] < bd or (gd[i] == bd and gi[i] < bi):
+                    bd, bi = gd[i], gi[i]
+            if bi >= 0 and bd < cur_d:
+                cur, cur_d = int(bi), bd
+            else:
+                return cur, cur_d
+
+    @staticmethod
+    def _hash(ident):
+        x = (ident * 0x9E3779B9) & 0xFFFFFFFF
+        x ^= x >> 16
+        return x & (HNSW_HASH - 1)
+
+    @classmethod
+    def _visited_insert(cls, tab, ident):
+        h = cls._hash(ident)
+        for p in range(HNSW_PROBES):
+            slot = (h + p) & (HNSW_HASH - 1)
+            v = tab.get(slot, 0)
+            if v == ident + 1:
+                return True
+            if v == 0:
+                tab[slot] = ident + 1
+                return False
+        return True  # cluster full: treated as visited
+
+    def _beam(self, u, entry, entry_d, level, ef):
+        tab = {}
+        entries = []  # (dist, id, expanded)
+        self._visited_insert(tab, entry)
+        entries.append([np.float32(entry_d), entry, False])
+
+        def key(e):
+            return (e[0], e[1] | (1 << 31) if e[2] else e[1])
+
+        while True:
+            entries.sort(key=lambda e: (e[0], (e[1] | (1 << 31)) if e[2] else e[1]))
+            del entries[ef:]  # Sel truncation to top-ef
+            pick = None
+            for e in entries:
+                if not e[2]:
+                    pick = e
+                    break
+            if pick is None:
+                return entries
+            pick[2] = True
+            for nid in self._nbrs(pick[1], level):
+                nid = int(nid)
+                dd = self._dist_q(u, nid)
+                if not self._visited_insert(tab, nid):
+                    entries.append([dd, nid, False])
+
+    def search(self, q, k, ef):
+        q = np.ascontiguousarray(q, dtype=np.float32)
+        nq = q.shape[0]
+        ef = max(ef, k)
+        g = self.g
+        D = np.full((nq, k), _FLT_MAX, dtype=np.float32)
+        I = np.full((nq, k), -1, dtype=np.int64)
+        for qi in range(nq):
+            u = ((q[qi] - self.vmin) - np.float32(0.5) * self.scale).astype(np.float32)
+            cur = int(g["entry"])
+            cur_d = self._dist_q(u, cur)
+            for l in range(int(g["maxlevel"]), 0, -1):
+                cur, cur_d = self._greedy(u, cur, cur_d, l)
+            res = self._beam(u, cur, cur_d, 0, ef)
+            res.sort(key=lambda e: (e[0], (e[1] | (1 << 31)) if e[2] else e[1]))
+            for j, e in enumerate(res[:k]):
+                D[qi, j] = e[0]
+                I[qi, j] = e[1]
+        return D, I

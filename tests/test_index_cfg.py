@@ -119,7 +119,24 @@ def test_infer_n_centroids_thresholds():
 
 
 def test_out_of_scope_builders_raise():
-    for t in ("hnswsq", "ivf_gpu"):
-        cfg = IndexCfg(index_builder_type=t, dim=16, metric="l2", centroids=4)
-        with pytest.raises(NotImplementedError):
-            resolve_engine_spec(cfg, 100)
+    # ivf_gpu stays out of scope (the faiss-GPU wrapper is what this
+    # engine replaces); hnswsq resolves since round 2
+    cfg = IndexCfg(index_builder_type="ivf_gpu", dim=16, metric="l2", centroids=4)
+    with pytest.raises(NotImplementedError):
+        resolve_engine_spec(cfg, 100)
+
+
+def test_builder_hnswsq():
+    # reference index.py:51-60: store_n/ef_construction from extra,
+    # L2 asserted
+    cfg = IndexCfg(index_builder_type="hnswsq", dim=32, metric="l2",
+                   nprobe=48, store_n=24, ef_construction=80)
+    spec = resolve_engine_spec(cfg, 100)
+    assert spec["type"] == "hnswsq" and spec["m"] == 24
+    assert spec["ef_construction"] == 80 and spec["nprobe"] == 48
+    cfg2 = IndexCfg(index_builder_type="hnswsq", dim=32, metric="l2")
+    spec2 = resolve_engine_spec(cfg2, 100)
+    assert spec2["m"] == 128 and spec2["ef_construction"] == 100  # defaults
+    with pytest.raises(AssertionError):  # reference asserts L2
+        resolve_engine_spec(
+            IndexCfg(index_builder_type="hnswsq", dim=32, metric="dot"), 100)
