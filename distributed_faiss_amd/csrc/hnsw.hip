@@ -27,8 +27,8 @@
 //    (faiss conventions).
 
 #define HNSW_MAXL 8
-#define HNSW_HASH 8192
-#define HNSW_PROBES 32
+#define HNSW_HASH 16384
+#define HNSW_PROBES 64
 
 // visited-hash helpers (LDS table of id+1, 0 = empty). Deterministic:
 // returns true if id was already present OR the probe window is full.

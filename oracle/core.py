@@ -822,8 +822,8 @@ def aggregate_results(shard_D, shard_I_meta, topk, maximize):
 # dump — oracle/__init__.py pinning notes).
 # ---------------------------------------------------------------------------
 
-HNSW_HASH = 8192
-HNSW_PROBES = 32
+HNSW_HASH = 16384
+HNSW_PROBES = 64
 HNSW_MAXL = 8
 _FLT_MAX = np.float32(3.402823466e+38)
 
