@@ -144,7 +144,8 @@ __device__ long long hnsw_greedy(const HnswGraph &g,
                                  const float *__restrict__ s,
                                  const uint8_t *const *__restrict__ codes,
                                  int rlog, int stride, int d, long long cur,
-                                 float &cur_d, int level, char *lds_scratch) {
+                                 float &cur_d, int level, char *lds_scratch,
+                                 int self_id = -1) {
   // lds_scratch: [blockDim/8] floats + ids for per-group minima
   float *gd = reinterpret_cast<float *>(lds_scratch);
   int *gi = reinterpret_cast<int *>(lds_scratch + (blockDim.x >> 3) * 4);
@@ -159,10 +160,12 @@ __device__ long long hnsw_greedy(const HnswGraph &g,
       int ci = c0 + grp;
       if (ci < cnt) {
         int nid = nb[ci];
-        float dd = hnsw_dist_q(u, s, codes, rlog, stride, d, nid, g8);
-        if (dd < best_d || (dd == best_d && nid < best_i)) {
-          best_d = dd;
-          best_i = nid;
+        if (nid != self_id) {
+          float dd = hnsw_dist_q(u, s, codes, rlog, stride, d, nid, g8);
+          if (dd < best_d || (dd == best_d && nid < best_i)) {
+            best_d = dd;
+            best_i = nid;
+          }
         }
       }
     }
@@ -203,7 +206,7 @@ __device__ void hnsw_beam(const HnswGraph &g, const float *__restrict__ u,
                           const uint8_t *const *__restrict__ codes, int rlog,
                           int stride, int d, long long entry, float entry_d,
                           int level, int ef, Sel &sel, unsigned *vis,
-                          float *nd_buf, int *ni_buf) {
+                          float *nd_buf, int *ni_buf, int self_id = -1) {
   // vis: HNSW_HASH LDS table; nd/ni_buf: blockDim/8 scratch
   for (int i = threadIdx.x; i < HNSW_HASH; i += blockDim.x) vis[i] = 0;
   sel_init(sel);
@@ -254,7 +257,8 @@ __device__ void hnsw_beam(const HnswGraph &g, const float *__restrict__ u,
         int lim = min(NG, cnt - c0);
         for (int i = 0; i < lim; ++i) {
           unsigned id = (unsigned)ni_buf[i];
-          if (!hnsw_visited_insert(vis, id)) sel_try(sel, nd_buf[i], id);
+          if ((int)id != self_id && !hnsw_visited_insert(vis, id))
+            sel_try(sel, nd_buf[i], id);
         }
       }
       __syncthreads();
@@ -343,7 +347,7 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_insert(
     const int *levels, int *nbr0, int *cnt0, const int *up_slot, int *nbrU,
     int *cntU, int deg0, int M, long long n_snap, long long entry,
     int entry_level, long long p0, int np, int efc, int *__restrict__ req,
-    int *__restrict__ req_cnt, int req_cap) {
+    int *__restrict__ req_cnt, int req_cap, int refine) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float *s_s = reinterpret_cast<float *>(smem);
   float *s_u = s_s + d;
@@ -369,14 +373,17 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_insert(
   int g8 = threadIdx.x & 7;
   int plevel = levels[p];
   if (n_snap == 0) return;  // first point: no links
+  int self_id = refine ? (int)p : -1;
+  if (refine && (long long)entry == p) return;  // entry re-links itself: skip
   long long cur = entry;
   float cur_d = hnsw_dist_q(s_u, s_s, codes, rlog, stride, d, cur, g8);
   __syncthreads();
   for (int l = entry_level; l > plevel; --l)
-    cur = hnsw_greedy(g, s_u, s_s, codes, rlog, stride, d, cur, cur_d, l, scr);
+    cur = hnsw_greedy(g, s_u, s_s, codes, rlog, stride, d, cur, cur_d, l, scr,
+                      self_id);
   for (int l = min(plevel, entry_level); l >= 0; --l) {
     hnsw_beam(g, s_u, s_s, codes, rlog, stride, d, cur, cur_d, l, efc, sel,
-              vis, nd_buf, ni_buf);
+              vis, nd_buf, ni_buf, self_id);
     // ---- faiss shrink heuristic: keep c if dist(p,c) < dist(c, kc)
     // for every kept kc (candidates in ascending distance) ----
     int res_cnt = *sel.cnt;
@@ -419,8 +426,9 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_insert(
       }
       __syncthreads();
     }
-    // write own links for level l
-    if (threadIdx.x == 0) {
+    // write own links for level l (refine keeps old links when the
+    // beam found nothing — never degrade connectivity)
+    if (threadIdx.x == 0 && !(refine && n_kept == 0)) {
       if (l == 0) {
         cnt0[p] = n_kept;
         for (int i = 0; i < n_kept; ++i) nbr0[p * (size_t)deg0 + i] = kept[i];
