@@ -6,8 +6,9 @@ from distributed_faiss_amd.hip_engine import HipEngine
 from oracle.core import OracleHNSWSearch
 
 def clustered(n, d, seed=0, centers=64, sigma=0.3):
+    crng = np.random.default_rng(1000)  # centers SHARED across calls
+    cent = crng.standard_normal((centers, d)).astype(np.float32) * 3.0
     rng = np.random.default_rng(seed)
-    cent = rng.standard_normal((centers, d)).astype(np.float32) * 3.0
     lbl = rng.integers(0, centers, n)
     return (cent[lbl] + sigma * rng.standard_normal((n, d)).astype(np.float32)).astype(np.float32)
 
@@ -30,6 +31,17 @@ dec = vmin[None, :] + (codes.astype(np.float32) + 0.5) * scale[None, :]
 q = clustered(200, d, seed=42)
 d2 = torch.cdist(torch.as_tensor(q).cuda(), torch.as_tensor(dec).cuda()) ** 2
 gt = torch.topk(d2, 10, largest=False).indices.cpu().numpy()
+# level-0 connectivity from the entry (host BFS on the dump)
+from collections import deque
+adj = g["nbr0"]; cnt = g["cnt0"]
+seen = np.zeros(n, dtype=bool)
+dq = deque([int(g["entry"])]); seen[int(g["entry"])] = True
+while dq:
+    u = dq.popleft()
+    for v in adj[u, :cnt[u]]:
+        if not seen[v]:
+            seen[v] = True; dq.append(int(v))
+print(f"level-0 reachable from entry: {seen.mean()*100:.1f}%")
 for ef in (32, 64, 128, 256):
     eng.nprobe = ef
     t0=time.time(); D, I = eng.search(q, 10); dt=time.time()-t0

@@ -27,8 +27,11 @@ from oracle.core import OracleHNSWSearch  # noqa: E402
 
 
 def _clustered(n, d, seed=0, centers=64, sigma=0.3):
+    # centers SHARED across calls (queries must come from the same
+    # mixture as the database, as the bench's gen_shard does)
+    crng = np.random.default_rng(1000)
+    cent = crng.standard_normal((centers, d)).astype(np.float32) * 3.0
     rng = np.random.default_rng(seed)
-    cent = rng.standard_normal((centers, d)).astype(np.float32) * 3.0
     lbl = rng.integers(0, centers, n)
     x = cent[lbl] + sigma * rng.standard_normal((n, d)).astype(np.float32)
     return x.astype(np.float32)
