@@ -1124,7 +1124,9 @@ static void hnsw_add(dfann_index *h, int64_t n, const float *x,
                        h->hn_cntU.as<int>(), deg0, M, w0, h->hnsw_entry,
                        h->hnsw_maxlevel, w0, (int)W, h->hnsw_efc,
                        h->hn_req.as<int>(), h->hn_reqcnt.as<int>(),
-                       (int)req_cap, 0);
+                       (int)req_cap, 0, (const int *)nullptr,
+                       (const int *)nullptr, (const int *)nullptr,
+                       (const int *)nullptr);
     HIP_CHECK(hipGetLastError());
     hnsw_apply_reqs(h, stream, lds_apply, req_cap);
     // entry update: highest new level (lowest id on ties) beats the old
@@ -1145,6 +1147,29 @@ static void hnsw_add(dfann_index *h, int64_t n, const float *x,
   // over the evolving-but-deterministic graph. spec "hnsw_refine" = 0
   // disables.
   if (json_int(h->spec_json, "hnsw_refine", 1) != 0) {
+    // freeze the pre-refine graph: refine blocks traverse the copy
+    // while own-link writes and reverse-link merges go to the live
+    // arrays — race-free and deterministic
+    DevBuf fz_nbr0, fz_cnt0, fz_nbrU, fz_cntU;
+    fz_nbr0.ensure((size_t)ntot * deg0 * 4);
+    fz_cnt0.ensure((size_t)ntot * 4);
+    HIP_CHECK(hipMemcpyAsync(fz_nbr0.p, h->hn_nbr0.p, (size_t)ntot * deg0 * 4,
+                             hipMemcpyDeviceToDevice, stream));
+    HIP_CHECK(hipMemcpyAsync(fz_cnt0.p, h->hn_cnt0.p, (size_t)ntot * 4,
+                             hipMemcpyDeviceToDevice, stream));
+    if (h->hnsw_nslots) {
+      fz_nbrU.ensure((size_t)h->hnsw_nslots * HNSW_MAXL * M * 4);
+      fz_cntU.ensure((size_t)h->hnsw_nslots * HNSW_MAXL * 4);
+      HIP_CHECK(hipMemcpyAsync(fz_nbrU.p, h->hn_nbrU.p,
+                               (size_t)h->hnsw_nslots * HNSW_MAXL * M * 4,
+                               hipMemcpyDeviceToDevice, stream));
+      HIP_CHECK(hipMemcpyAsync(fz_cntU.p, h->hn_cntU.p,
+                               (size_t)h->hnsw_nslots * HNSW_MAXL * 4,
+                               hipMemcpyDeviceToDevice, stream));
+    } else {
+      fz_nbrU.ensure(16);
+      fz_cntU.ensure(16);
+    }
     int64_t r0 = n0;
     while (r0 < ntot) {
       int64_t W = std::min<int64_t>(WMAX, ntot - r0);
@@ -1162,7 +1187,9 @@ static void hnsw_add(dfann_index *h, int64_t n, const float *x,
                          h->hn_cntU.as<int>(), deg0, M, ntot, h->hnsw_entry,
                          h->hnsw_maxlevel, r0, (int)W, h->hnsw_efc,
                          h->hn_req.as<int>(), h->hn_reqcnt.as<int>(),
-                         (int)req_cap, 1);
+                         (int)req_cap, 1, fz_nbr0.as<const int>(),
+                         fz_cnt0.as<const int>(), fz_nbrU.as<const int>(),
+                         fz_cntU.as<const int>());
       HIP_CHECK(hipGetLastError());
       hnsw_apply_reqs(h, stream, lds_apply, req_cap);
       r0 += W;

@@ -347,7 +347,9 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_insert(
     const int *levels, int *nbr0, int *cnt0, const int *up_slot, int *nbrU,
     int *cntU, int deg0, int M, long long n_snap, long long entry,
     int entry_level, long long p0, int np, int efc, int *__restrict__ req,
-    int *__restrict__ req_cnt, int req_cap, int refine) {
+    int *__restrict__ req_cnt, int req_cap, int refine,
+    const int *r_nbr0, const int *r_cnt0, const int *r_nbrU,
+    const int *r_cntU) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float *s_s = reinterpret_cast<float *>(smem);
   float *s_u = s_s + d;
@@ -367,8 +369,13 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_insert(
     s_u[t] = up[(size_t)blockIdx.x * d + t];
   }
   __syncthreads();
-  HnswGraph g{levels, (const int *)nbr0, (const int *)cnt0, up_slot,
-              (const int *)nbrU, (const int *)cntU, deg0, M};
+  // refine traverses a FROZEN copy of the graph (r_*): a refine block
+  // rewrites p's adjacency while other blocks may traverse p — reading
+  // the live arrays would race into never-initialized slots
+  HnswGraph g{levels, r_nbr0 ? r_nbr0 : (const int *)nbr0,
+              r_cnt0 ? r_cnt0 : (const int *)cnt0, up_slot,
+              r_nbrU ? r_nbrU : (const int *)nbrU,
+              r_cntU ? r_cntU : (const int *)cntU, deg0, M};
   Sel sel = sel_carve(selbase);
   int g8 = threadIdx.x & 7;
   int plevel = levels[p];
