@@ -1076,12 +1076,19 @@ static void hnsw_add(dfann_index *h, int64_t n, const float *x,
                            h->h_upslot.data() + n0, (size_t)n * 4,
                            hipMemcpyHostToDevice, stream));
   h->hn_nbr0.grow_keep((size_t)ntot * deg0 * 4, (size_t)n0 * deg0 * 4);
+  // -1-fill fresh adjacency: slots beyond each cnt stay deterministic
+  // (dump comparisons, persistence) instead of hipMalloc garbage
+  HIP_CHECK(hipMemsetAsync(h->hn_nbr0.as<int>() + (size_t)n0 * deg0, 0xFF,
+                           (size_t)n * deg0 * 4, stream));
   h->hn_cnt0.grow_keep((size_t)ntot * 4, (size_t)n0 * 4);
   HIP_CHECK(hipMemsetAsync(h->hn_cnt0.as<int>() + n0, 0, (size_t)n * 4,
                            stream));
   if (h->hnsw_nslots > old_nslots) {
     h->hn_nbrU.grow_keep((size_t)h->hnsw_nslots * HNSW_MAXL * M * 4,
                          (size_t)old_nslots * HNSW_MAXL * M * 4);
+    HIP_CHECK(hipMemsetAsync(
+        h->hn_nbrU.as<int>() + (size_t)old_nslots * HNSW_MAXL * M, 0xFF,
+        (size_t)(h->hnsw_nslots - old_nslots) * HNSW_MAXL * M * 4, stream));
     h->hn_cntU.grow_keep((size_t)h->hnsw_nslots * HNSW_MAXL * 4,
                          (size_t)old_nslots * HNSW_MAXL * 4);
     HIP_CHECK(hipMemsetAsync(
