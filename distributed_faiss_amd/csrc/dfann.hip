@@ -1311,6 +1311,11 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
           if (const char *e = getenv("DFANN_SCAN_NT"))
             if (atoi(e) == 1)
               kern = ip ? k_scan_sq8_ip_rk_nt : k_scan_sq8_l2_rk_nt;
+        // DFANN_SCAN_U8=1: 8 rows in flight per wave group (A/B)
+        if (rk)
+          if (const char *e = getenv("DFANN_SCAN_U8"))
+            if (atoi(e) == 1)
+              kern = ip ? k_scan_sq8_ip_rk_u8 : k_scan_sq8_l2_rk_u8;
       } else {
         fam_floats = h->d;
         kern = rk ? (ip ? k_scan_sqf_ip_rk : k_scan_sqf_l2_rk)

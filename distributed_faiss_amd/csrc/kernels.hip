@@ -1142,7 +1142,7 @@ __device__ __forceinline__ float scan_row_dist(const uint8_t *__restrict__ cp,
 }
 
 template <int FAM, bool IS_IP, bool REGSEL, bool PRE = false,
-          bool GLUT = false, bool L16 = false, bool NT = false>
+          bool GLUT = false, bool L16 = false, bool NT = false, int UR = 4>
 __device__ void ivf_scan_body(
     const float *__restrict__ q, const float *__restrict__ cent,
     const float *__restrict__ cb, const float *__restrict__ sq_vmin,
@@ -1352,15 +1352,15 @@ __device__ void ivf_scan_body(
       int g8 = threadIdx.x & 7, grp = threadIdx.x >> 3;  // 32 row-groups
       const bool onechunk = d <= 128;  // lane covers one 16-B chunk
       const int NG8 = blockDim.x >> 3;  // row groups per block
-      for (long long base = s0; base < s1; base += (long long)NG8 * 4) {
-        if (!REGSEL) sel_guard(s, k, NG8 * 4);
+      for (long long base = s0; base < s1; base += (long long)NG8 * UR) {
+        if (!REGSEL) sel_guard(s, k, NG8 * UR);
         if (onechunk) {
-          // issue all 4 row-set loads up front: 4 independent HBM
+          // issue all UR row-set loads up front: UR independent HBM
           // requests in flight per wave instead of 1 (latency hiding)
-          uint4 wv4[4];
-          bool val4[4];
+          uint4 wv4[UR];
+          bool val4[UR];
 #pragma unroll
-          for (int u = 0; u < 4; ++u) {
+          for (int u = 0; u < UR; ++u) {
             long long pos = base + (long long)u * NG8 + grp;
             val4[u] = pos < s1;
             int t0 = g8 * 16;
@@ -1374,7 +1374,7 @@ __device__ void ivf_scan_body(
                          : uint4{0, 0, 0, 0};
           }
 #pragma unroll
-          for (int u = 0; u < 4; ++u) {
+          for (int u = 0; u < UR; ++u) {
             long long pos = base + (long long)u * NG8 + grp;
             float part = 0.f;
             int t0 = g8 * 16;
@@ -1408,7 +1408,7 @@ __device__ void ivf_scan_body(
           }
         } else {
 #pragma unroll
-          for (int u = 0; u < 4; ++u) {
+          for (int u = 0; u < UR; ++u) {
             long long pos = base + (long long)u * NG8 + grp;
             bool valid = pos < s1;
             float part = 0.f;
@@ -1894,6 +1894,22 @@ extern "C" __global__ __launch_bounds__(256) void k_pq_lut_f16(
   }
 INSTANTIATE_SCAN_NT(k_scan_sq8_l2_rk_nt, 2, false, true)
 INSTANTIATE_SCAN_NT(k_scan_sq8_ip_rk_nt, 2, true, true)
+
+#define INSTANTIATE_SCAN_U8(NAME, FAM, IS_IP, REGSEL)                          \
+  extern "C" __global__ __launch_bounds__(512) void NAME(                      \
+      const float *q, const float *cent, const float *cb,                      \
+      const float *sq_vmin, const float *sq_scale, const int *probes,          \
+      const float *keys, const uint8_t *const *codes, const int64_t *off,      \
+      int nq,                                                                  \
+      int nprobe, int d, int m, int dsub, int k, int stride, int rlog,         \
+      float *cand_d, unsigned *cand_p, int fam_floats, int fan) {              \
+    ivf_scan_body<FAM, IS_IP, REGSEL, false, false, false, false, 8>(          \
+        q, cent, cb, sq_vmin, sq_scale, probes, keys, codes, off, nq, nprobe,  \
+        d, m, dsub, k, stride, rlog, cand_d, cand_p, fam_floats, nullptr,      \
+        nullptr, nullptr, fan);                                                \
+  }
+INSTANTIATE_SCAN_U8(k_scan_sq8_l2_rk_u8, 2, false, true)
+INSTANTIATE_SCAN_U8(k_scan_sq8_ip_rk_u8, 2, true, true)
 
 INSTANTIATE_SCAN(k_scan_ivfflat_l2_rk, 1, false, true)
 INSTANTIATE_SCAN(k_scan_ivfflat_ip_rk, 1, true, true)
