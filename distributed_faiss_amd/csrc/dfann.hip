@@ -1097,7 +1097,8 @@ static void hnsw_add(dfann_index *h, int64_t n, const float *x,
   }
   // 4) wave insertion over frozen snapshots
   const int64_t WMAX = 4096;
-  const int64_t req_cap = WMAX * M * (HNSW_MAXL + 1);
+  // x2: the refine pass queues both directions per kept link
+  const int64_t req_cap = 2 * WMAX * M * (HNSW_MAXL + 1);
   h->hn_req.ensure((size_t)req_cap * 16);
   h->hn_reqcnt.ensure(4);
   h->hn_u.ensure((size_t)WMAX * h->d * 4);

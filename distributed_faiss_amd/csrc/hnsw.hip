@@ -433,26 +433,45 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_insert(
       }
       __syncthreads();
     }
-    // write own links for level l (refine keeps old links when the
-    // beam found nothing — never degrade connectivity)
-    if (threadIdx.x == 0 && !(refine && n_kept == 0)) {
-      if (l == 0) {
-        cnt0[p] = n_kept;
-        for (int i = 0; i < n_kept; ++i) nbr0[p * (size_t)deg0 + i] = kept[i];
+    // initial insertion: write own links directly (this point's rows
+    // are untouched by other blocks) + queue reverse requests.
+    // REFINE: do NOT replace own links — the early waves' long-range
+    // links are the graph's inter-cluster bridges (measured: replacing
+    // them caps recall ~0.6 at 1M); instead queue BOTH directions
+    // through the sorted merge+prune (k_hnsw_apply dedups).
+    if (threadIdx.x == 0 && n_kept > 0) {
+      if (!refine) {
+        if (l == 0) {
+          cnt0[p] = n_kept;
+          for (int i = 0; i < n_kept; ++i) nbr0[p * (size_t)deg0 + i] = kept[i];
+        } else {
+          int slot = up_slot[p];
+          cntU[(size_t)slot * HNSW_MAXL + (l - 1)] = n_kept;
+          for (int i = 0; i < n_kept; ++i)
+            nbrU[((size_t)slot * HNSW_MAXL + (l - 1)) * M + i] = kept[i];
+        }
+        int base = atomicAdd(req_cnt, n_kept);
+        for (int i = 0; i < n_kept && base + i < req_cap; ++i) {
+          int *r = req + (size_t)(base + i) * 4;
+          r[0] = kept[i];
+          r[1] = (int)p;
+          r[2] = l;
+          reinterpret_cast<float *>(r)[3] = kept_d[i];
+        }
       } else {
-        int slot = up_slot[p];
-        cntU[(size_t)slot * HNSW_MAXL + (l - 1)] = n_kept;
-        for (int i = 0; i < n_kept; ++i)
-          nbrU[((size_t)slot * HNSW_MAXL + (l - 1)) * M + i] = kept[i];
-      }
-      // queue reverse requests (dst, src, level, dist)
-      int base = atomicAdd(req_cnt, n_kept);
-      for (int i = 0; i < n_kept && base + i < req_cap; ++i) {
-        int *r = req + (size_t)(base + i) * 4;
-        r[0] = kept[i];
-        r[1] = (int)p;
-        r[2] = l;
-        reinterpret_cast<float *>(r)[3] = kept_d[i];
+        int base = atomicAdd(req_cnt, 2 * n_kept);
+        for (int i = 0; i < n_kept && base + 2 * i + 1 < req_cap; ++i) {
+          int *r = req + (size_t)(base + 2 * i) * 4;
+          r[0] = kept[i];
+          r[1] = (int)p;
+          r[2] = l;
+          reinterpret_cast<float *>(r)[3] = kept_d[i];
+          int *r2 = req + (size_t)(base + 2 * i + 1) * 4;
+          r2[0] = (int)p;
+          r2[1] = kept[i];
+          r2[2] = l;
+          reinterpret_cast<float *>(r2)[3] = kept_d[i];
+        }
       }
     }
     __syncthreads();
@@ -504,11 +523,18 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_apply(
   int old_cnt = *cnt_p;
   int inc = r1 - r0;
   if (old_cnt + inc <= cap) {
-    // fast path: append in sorted request order
+    // fast path: append in sorted request order, skipping srcs already
+    // linked (refine re-proposes existing links)
     if (threadIdx.x == 0) {
-      for (int i = 0; i < inc; ++i)
-        nbrs[old_cnt + i] = req[(size_t)(r0 + i) * 4 + 1];
-      *cnt_p = old_cnt + inc;
+      int w = old_cnt;
+      for (int i = 0; i < inc; ++i) {
+        int src = req[(size_t)(r0 + i) * 4 + 1];
+        bool dup = false;
+        for (int j = 0; j < w; ++j)
+          if (nbrs[j] == src) { dup = true; break; }
+        if (!dup) nbrs[w++] = src;
+      }
+      *cnt_p = w;
     }
     return;
   }
@@ -563,6 +589,11 @@ extern "C" __global__ __launch_bounds__(256) void k_hnsw_apply(
   for (int ci = 0; ci < total && n_kept < cap; ++ci) {
     int id = cid[ci];
     float dd = cdist[ci];
+    if (threadIdx.x == 0) {
+      for (int j = 0; j < n_kept; ++j)
+        if (kept[j] == id) { drop_marker = 0; break; }
+    }
+    __syncthreads();
     for (int k0 = 0; k0 < n_kept; k0 += NG) {
       int ki = k0 + grp;
       float kd = DFANN_FLT_MAX;

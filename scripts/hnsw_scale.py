@@ -12,10 +12,14 @@ def clustered(n, d, seed=0, centers=4096, sigma=0.3):
     lbl = rng.integers(0, centers, n)
     return (cent[lbl] + sigma * rng.standard_normal((n, d)).astype(np.float32)).astype(np.float32)
 
-n, d, M, efc = 1_000_000, 64, 32, 100
+import os
+n = int(os.environ.get("HN_N", "1000000"))
+refine = int(os.environ.get("HN_REFINE", "1"))
+d, M, efc = 64, 32, 100
 xb = clustered(n, d, seed=7)
 spec = {"type": "hnswsq", "dim": d, "metric": 1, "m": M,
-        "ef_construction": efc, "nprobe": 64, "seed": 11}
+        "ef_construction": efc, "nprobe": 64, "seed": 11,
+        "hnsw_refine": refine}
 eng = HipEngine(spec=spec)
 eng.train(xb[:100_000])
 t0 = time.time()
