@@ -37,6 +37,15 @@
 #   - ties everywhere break to the lower id (faiss heap behavior for
 #     within-list scans; cross-list tie order in faiss is heap-order
 #     dependent and unpinned).
+#
+# hnswsq tier (round 2): the SEARCH side is restated
+# op-for-op (core.OracleHNSWSearch) and runs over the ENGINE'S OWN dumped
+# graph — engine search results are bitwise equal to it
+# (tests/test_hnsw.py). The BUILD side is NOT restated: the engine's
+# batched wave insertion is a documented deviation from faiss's
+# sequential insertion (parity unpinned there by construction); the
+# build's pinned contracts are determinism (identical graph dumps) and
+# recall property gates.
 
 from .core import (  # noqa: F401
     METRIC_INNER_PRODUCT,
