@@ -649,6 +649,12 @@ static dfann_index *create_from_spec(const std::string &js) {
     if (h->metric != M_L2)
       { delete h; throw std::runtime_error("hnswsq requires L2 (ref index.py:52)"); }
     if (h->m <= 0) h->m = 128;  // reference store_n default
+    if (h->m > 256) {
+      delete h;
+      throw std::runtime_error(
+          "hnswsq store_n > 256 unsupported (level-0 degree cap 2M is "
+          "sized 512 in the merge kernels)");
+    }
     h->hnsw_efc = (int)json_int(js, "ef_construction", 100);
     if (h->hnsw_efc < 1) h->hnsw_efc = 1;
     if (h->hnsw_efc > 512) h->hnsw_efc = 512;
