@@ -1376,13 +1376,13 @@ static void scan_and_merge(dfann_index *h, int64_t nq, const float *q,
   // (+8% over 384); SQ8 10M 128 threads +8%. The LDS-buffer selection
   // path keeps 256 (Sel capacity is sized for it).
   unsigned scan_bs = 256;
-  // round-2 addition: tiny-LDS scans (SQ8/SQfp16/IVF-Flat staging
-  // regions ~1 KB) drop to 64-thread blocks — measured +4% on the
-  // config-5 125M SQ8 scan (2075 vs 1991 GB/s, gpurun_out/r2j_*): even
-  // more independent per-CU phases than the r1 128-thread finding
-  if (rk) scan_bs = lds <= 4 * 1024 ? 64
-                    : lds <= 16 * 1024 ? 128
-                    : (lds <= 32 * 1024 ? 256 : 512);
+  // round-2 addition: SQ scans drop to 64-thread blocks — measured +4%
+  // on the config-5 125M SQ8 scan (2075 vs 1991 GB/s, gpurun_out/r2j_*)
+  // and +6% at 10M; even more independent per-CU phases than the r1
+  // 128-thread finding. IVF-Flat measured −5% at 64 (r2z_ivfflat), so
+  // the drop is SQ-only.
+  if (rk) scan_bs = lds <= 16 * 1024 ? 128 : (lds <= 32 * 1024 ? 256 : 512);
+  if (rk && h->type == T_IVFSQ && lds <= 4 * 1024) scan_bs = 64;
   if (const char *e = getenv("DFANN_SCAN_BS"))  // experiment override
     if (int v = atoi(e)) scan_bs = (unsigned)((v / 64) * 64);
   h->ws3.ensure((size_t)nq * nprobe * fan * k * 4);
