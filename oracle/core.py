@@ -929,10 +929,6 @@ class OracleHNSWSearch:
         entries = []  # (dist, id, expanded)
         self._visited_insert(tab, entry)
         entries.append([np.float32(entry_d), entry, False])
-
-        def key(e):
-            return (e[0], e[1] | (1 << 31) if e[2] else e[1])
-
         while True:
             entries.sort(key=lambda e: (e[0], (e[1] | (1 << 31)) if e[2] else e[1]))
             del entries[ef:]  # Sel truncation to top-ef
