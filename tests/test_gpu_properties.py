@@ -139,3 +139,92 @@ def test_incremental_merge_equals_bulk():
         Db, Ib = b.search(q, k)
         np.testing.assert_array_equal(Ia, Ib)
         np.testing.assert_array_equal(Da, Db)
+
+
+def test_engine_caps_and_filtered_overfetch(tmp_path):
+    # include/dfann.h Limits: k <= 512 errors; nprobe > 512 clamps with a
+    # warning; the filtered over-fetch clamps instead of raising
+    # 
+    import numpy as np
+
+    from distributed_faiss_amd import IndexCfg, IndexClient, IndexServer
+    from distributed_faiss_amd.hip_engine import HipEngine, HipProvider
+
+    rng = np.random.default_rng(9)
+    d, n = 32, 30_000
+    xb = rng.standard_normal((n, d), dtype=np.float32)
+    spec = {"type": "ivfpq", "dim": d, "metric": 1, "nlist": 600, "m": 8,
+            "nbits": 8, "nprobe": 4, "seed": 2}
+    eng = HipEngine(spec=spec)
+    eng.train(xb[:20_000])
+    eng.add(xb)
+    q = rng.standard_normal((8, d), dtype=np.float32)
+    # k boundary: 512 works, 513 raises the documented error
+    D, I = eng.search(q, 512)
+    assert D.shape == (8, 512)
+    with pytest.raises(RuntimeError, match="k > 512"):
+        eng.search(q, 513)
+    # nprobe above the cap: clamped (warns once), results still valid
+    eng.nprobe = 600
+    D2, I2 = eng.search(q, 10)
+    assert (np.diff(D2, axis=1) >= 0).all()
+    eng.nprobe = 599  # nlist-bounded clamp still sees every list
+    D3, _ = eng.search(q, 10)
+    np.testing.assert_array_equal(D2, D3)
+
+    # client: filtered search with top_k >= 171 must not raise (the 3x
+    # over-fetch is clamped to the engine cap)
+    prov = HipProvider()
+    srv = IndexServer(0, str(tmp_path), provider=prov)
+    cli = IndexClient(servers=[srv])
+    cfg = IndexCfg(index_builder_type="flat", dim=d, metric="dot",
+                   train_num=100)
+    cli.create_index("c", cfg)
+    cli.cfg = cfg
+    cli.add_index_data("c", xb[:5000], [("a" if i % 2 else "b", i)
+                                        for i in range(5000)],
+                       train_async_if_triggered=False)
+    import time
+
+    from distributed_faiss_amd import IndexState
+
+    for _ in range(200):
+        if (cli.get_state("c") == IndexState.TRAINED
+                and cli.get_ntotal("c") == 5000):
+            break
+        time.sleep(0.05)
+    scores, meta = cli.search_with_filter(q, 180, "c", filter_pos=0,
+                                          filter_value="a")
+    assert len(meta) == 8
+    for row in meta:
+        for m in row:
+            assert m[0] != "a"
+
+
+def test_hnsw_tiny_and_incremental(tmp_path):
+    # hnsw edges: n=1 search; add-after-save/load keeps working
+    import numpy as np
+
+    from distributed_faiss_amd.hip_engine import HipEngine, HipProvider
+
+    rng = np.random.default_rng(4)
+    d = 16
+    spec = {"type": "hnswsq", "dim": d, "metric": 1, "m": 8,
+            "ef_construction": 20, "nprobe": 10, "seed": 5}
+    eng = HipEngine(spec=spec)
+    eng.train(rng.standard_normal((500, d), dtype=np.float32))
+    one = rng.standard_normal((1, d), dtype=np.float32)
+    eng.add(one)
+    D, I = eng.search(rng.standard_normal((3, d), dtype=np.float32), 2)
+    assert (I[:, 0] == 0).all() and (I[:, 1] == -1).all()
+    # grow incrementally, save/load, grow again
+    eng.add(rng.standard_normal((800, d), dtype=np.float32))
+    p = str(tmp_path / "t.dfann")
+    eng.save(p)
+    e2 = HipProvider().load(p)
+    assert e2.ntotal == 801
+    e2.add(rng.standard_normal((300, d), dtype=np.float32))
+    q = rng.standard_normal((5, d), dtype=np.float32)
+    D2, I2 = e2.search(q, 5)
+    assert (I2 >= 0).all() and (I2 < 1101).all()
+    assert (np.diff(D2, axis=1) >= 0).all()
