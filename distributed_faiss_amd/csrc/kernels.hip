@@ -626,16 +626,24 @@ extern "C" __global__ __launch_bounds__(256) void k_topk_rows(
   const float *kp = keys + row * ldk;
   const int BS = blockDim.x;
   if ((ldk & 3) == 0) {
-    // float4 row reads (see k_topk_rows_rk) — Sel is order-invariant
+    // float4 row reads (see k_topk_rows_rk) — Sel is order-invariant.
+    // Appends are split into TWO guarded half-batches of 2*BS: a single
+    // 4*BS batch can exceed the Sel headroom (SEL_CAP - k) at large k —
+    // at k=512 the buffer overflowed its LDS carve (caught by
+    // test_engine_caps_and_filtered_overfetch).
     const float4 *kp4 = reinterpret_cast<const float4 *>(kp);
     long long c4n = cols >> 2;
     for (long long c0 = 0; c0 < c4n; c0 += BS) {
-      sel_guard(s, k, 4 * BS);
       long long c4 = c0 + threadIdx.x;
-      if (c4 < c4n) {
-        float4 v4 = kp4[c4];
+      bool v = c4 < c4n;
+      float4 v4 = v ? kp4[c4] : float4{0.f, 0.f, 0.f, 0.f};
+      sel_guard(s, k, 2 * BS);
+      if (v) {
         sel_try(s, v4.x, (unsigned)(c4 * 4 + 0) + base);
         sel_try(s, v4.y, (unsigned)(c4 * 4 + 1) + base);
+      }
+      sel_guard(s, k, 2 * BS);
+      if (v) {
         sel_try(s, v4.z, (unsigned)(c4 * 4 + 2) + base);
         sel_try(s, v4.w, (unsigned)(c4 * 4 + 3) + base);
       }
